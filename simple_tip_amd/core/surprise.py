@@ -445,14 +445,21 @@ class DSA(SA):
         target_ats = target_ats.to(dev, self.train_activations.dtype)
         target_pred = target_pred.to(dev)
         dsa = torch.empty(target_pred.shape[0], dtype=target_ats.dtype, device=dev)
-        for label in range(self.num_classes):
+        for label in torch.unique(target_pred).tolist():
             sel = target_pred == label
-            if not bool(sel.any()):
-                continue
             same = self.train_activations[self.train_predictions == label]
             other = self.train_activations[self.train_predictions != label]
+            if same.shape[0] == 0:
+                # predicted class absent from the (subsampled) training
+                # predictions: maximally surprising. (The reference would
+                # crash here; only reachable with tiny/degenerate data.)
+                dsa[sel] = float("inf")
+                continue
             samples = target_ats[sel]
             dist_a, closest_idx = ops.rowmin_l2(samples, same)
+            if other.shape[0] == 0:
+                dsa[sel] = 0.0  # single-class training set: no contrast
+                continue
             closest_ats = same[closest_idx]
             dist_b, _ = ops.rowmin_l2(closest_ats, other)
             dsa[sel] = dist_a / dist_b

@@ -1,0 +1,229 @@
+"""Abstract case study: owns data, model factory, training loop and the four
+experiment phases (train / test_prio / active_learning / at_collection).
+
+Capability parity with reference src/dnn_test_prio/case_study.py:13-144.
+The ensemble is a spawn process pool over model ids (engine/ensemble.py);
+training runs in bf16 autocast on the GPU.
+"""
+
+import logging
+import os
+from functools import partial
+from typing import Callable, List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+from ..config import MAX_NUM_MODELS, StudyConfig
+from ..engine import ensemble
+from ..engine import eval_prioritization, eval_active_learning, activation_persistor
+from ..models.base import TapModel
+
+logger = logging.getLogger(__name__)
+
+
+def default_device() -> torch.device:
+    return torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
+
+
+def train_classifier(
+    model: TapModel,
+    x: np.ndarray,
+    y: np.ndarray,
+    epochs: int,
+    batch_size: int,
+    lr: float = 1e-3,
+    device: Optional[torch.device] = None,
+    validation_split: float = 0.1,
+    seed: Optional[int] = None,
+) -> TapModel:
+    """Adam + cross-entropy training loop (bf16 autocast on GPU).
+
+    Plays the role of the reference's keras ``model.fit`` calls
+    (e.g. case_study_mnist.py:68)."""
+    device = device or default_device()
+    if seed is not None:
+        torch.manual_seed(seed)
+    model = model.to(device)
+    model.train()
+    opt = torch.optim.Adam(model.parameters(), lr=lr)
+    loss_fn = nn.CrossEntropyLoss()
+    n = x.shape[0]
+    n_train = int(n * (1 - validation_split)) if validation_split else n
+    xt = torch.as_tensor(np.ascontiguousarray(x))
+    if xt.dtype == torch.float64:
+        xt = xt.float()
+    yt = torch.as_tensor(np.asarray(y).reshape(-1)).long()
+    use_amp = device.type == "cuda"
+    for epoch in range(epochs):
+        perm = torch.randperm(n_train)
+        total, correct, loss_sum = 0, 0, 0.0
+        for s in range(0, n_train, batch_size):
+            idx = perm[s : s + batch_size]
+            xb = xt[idx].to(device, non_blocking=True)
+            yb = yt[idx].to(device, non_blocking=True)
+            opt.zero_grad(set_to_none=True)
+            with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_amp):
+                logits = model(xb)
+                loss = loss_fn(logits.float(), yb)
+            loss.backward()
+            opt.step()
+            loss_sum += float(loss.detach()) * len(idx)
+            correct += int((logits.argmax(dim=1) == yb).sum())
+            total += len(idx)
+        logger.info(
+            "epoch %d/%d loss %.4f acc %.3f", epoch + 1, epochs,
+            loss_sum / max(total, 1), correct / max(total, 1),
+        )
+    model.eval()
+    return model
+
+
+class CaseStudy:
+    """Base class; subclasses provide config, model factory and data."""
+
+    config: StudyConfig
+
+    def __init__(self, scale: Optional[float] = None, device=None):
+        # scale < 1 shrinks dataset sizes/epochs (CPU tests, smoke runs)
+        self.scale = scale if scale is not None else float(
+            os.environ.get("TIP_SCALE", "1.0")
+        )
+        self.device = device or default_device()
+
+    # -- to be provided by subclasses -----------------------------------
+    def build_model(self) -> TapModel:
+        raise NotImplementedError
+
+    def load_datasets(self) -> Tuple[
+        Tuple[np.ndarray, np.ndarray],
+        Tuple[np.ndarray, np.ndarray],
+        Tuple[np.ndarray, np.ndarray],
+    ]:
+        """((train_x, train_y), (nominal_x, nominal_y), (ood_x, ood_y))."""
+        raise NotImplementedError
+
+    # -- derived sizes ---------------------------------------------------
+    def _n(self, n: int) -> int:
+        return max(64, int(n * self.scale))
+
+    def _epochs(self) -> int:
+        return max(1, int(round(self.config.epochs * min(1.0, self.scale * 2))))
+
+    def training_process(self, x: np.ndarray, y: np.ndarray) -> TapModel:
+        """Train a fresh model on (x, y) — used by train and retrain."""
+        return train_classifier(
+            self.build_model(),
+            x,
+            y,
+            epochs=self._epochs(),
+            batch_size=self.config.train_batch,
+            device=self.device,
+        )
+
+    # -- phases ----------------------------------------------------------
+    def train(self, model_ids: List[int], num_processes: int = 0, context=None) -> None:
+        """Train and persist one model per id."""
+        assert all(0 <= i < MAX_NUM_MODELS for i in model_ids)
+        ensemble.run_tasks(
+            partial(_train_task, type(self), self.scale), model_ids, num_processes
+        )
+
+    def run_prio_eval(self, model_ids: List[int], num_processes: int = 0, context=None) -> None:
+        ensemble.run_tasks(
+            partial(_prio_task, type(self), self.scale), model_ids, num_processes
+        )
+
+    def run_active_learning_eval(
+        self, model_ids: List[int], num_processes: int = 0, context=None
+    ) -> None:
+        ensemble.run_tasks(
+            partial(_active_task, type(self), self.scale), model_ids, num_processes
+        )
+
+    def collect_activations(
+        self, model_ids: List[int], num_processes: int = 0, context=None
+    ) -> None:
+        ensemble.run_tasks(
+            partial(_at_collection_task, type(self), self.scale), model_ids, num_processes
+        )
+
+    # -- helpers ---------------------------------------------------------
+    def _load_model(self, model_id: int) -> TapModel:
+        return ensemble.load_model(
+            self.config.name, model_id, self.build_model, device=self.device
+        )
+
+
+# -- module-level task functions (picklable for the spawn pool) -------------
+
+def _train_task(study_cls, scale, model_id: int):
+    study = study_cls(scale=scale)
+    (train_x, train_y), _, _ = study.load_datasets()
+    torch.manual_seed(model_id)
+    np.random.seed(model_id)
+    model = study.training_process(train_x, train_y)
+    ensemble.save_model(study.config.name, model_id, model)
+    logger.info("trained and saved %s model %d", study.config.name, model_id)
+
+
+def _prio_task(study_cls, scale, model_id: int):
+    study = study_cls(scale=scale)
+    (train_x, _), (nom_x, nom_y), (ood_x, ood_y) = study.load_datasets()
+    model = study._load_model(model_id)
+    eval_prioritization.evaluate(
+        model_id=model_id,
+        case_study=study.config.name,
+        model=model,
+        training_dataset=train_x,
+        nominal_test_dataset=nom_x,
+        nominal_test_labels=nom_y,
+        ood_test_dataset=ood_x,
+        ood_test_labels=ood_y,
+        nc_activation_layers=study.config.nc_layers,
+        sa_activation_layers=study.config.sa_layers,
+        dsa_badge_size=study.config.dsa_badge_size,
+        device=study.device,
+        predict_batch=study.config.predict_batch,
+    )
+
+
+def _active_task(study_cls, scale, model_id: int):
+    study = study_cls(scale=scale)
+    (train_x, train_y), (nom_x, nom_y), (ood_x, ood_y) = study.load_datasets()
+    model = study._load_model(model_id)
+    eval_active_learning.evaluate(
+        model_id=model_id,
+        case_study=study.config.name,
+        model=model,
+        train_x=train_x,
+        train_y=train_y,
+        nominal_test_x=nom_x,
+        nominal_test_labels=nom_y,
+        ood_test_x=ood_x,
+        ood_test_labels=ood_y,
+        nc_activation_layers=study.config.nc_layers,
+        sa_activation_layers=study.config.sa_layers,
+        training_process=study.training_process,
+        observed_share=study.config.observed_share,
+        num_selected=max(8, int(study.config.num_selected * study.scale)),
+        num_classes=study.config.num_classes,
+        dsa_badge_size=study.config.dsa_badge_size,
+        device=study.device,
+        predict_batch=study.config.predict_batch,
+    )
+
+
+def _at_collection_task(study_cls, scale, model_id: int):
+    study = study_cls(scale=scale)
+    (train_x, _), (nom_x, _), (ood_x, _) = study.load_datasets()
+    model = study._load_model(model_id)
+    activation_persistor.persist(
+        case_study=study.config.name,
+        model_id=model_id,
+        model=model,
+        datasets={"train": train_x, "nominal": nom_x, "ood": ood_x},
+        num_layers=len(model.layers),
+        device=study.device,
+    )

@@ -1,3 +1,12 @@
+import os
+import tempfile
+
+# Route the /assets artifact fabric to a scratch dir BEFORE any
+# simple_tip_amd import resolves the layout.
+os.environ.setdefault(
+    "TIP_ASSETS_DIR", tempfile.mkdtemp(prefix="tip_assets_test_")
+)
+
 import pytest
 
 
