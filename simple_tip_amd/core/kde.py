@@ -110,6 +110,33 @@ class StableGaussianKDE:
             y = y.to(dtype=dtype or y.dtype, device=device or y.device)
         return y
 
+    def device_state(self, device, dtype=torch.float32):
+        """Cache (L^-T, whitened train, const) on the device for the fast
+        path: whitening becomes one fp32 GEMM feeding the MFMA KDE kernel."""
+        key = (str(device), dtype)
+        cache = getattr(self, "_dev_cache", None)
+        if cache is None:
+            cache = self._dev_cache = {}
+        if key not in cache:
+            eye = torch.eye(self.d, dtype=torch.float64)
+            linv = torch.linalg.solve_triangular(self.cho_l, eye, upper=False)
+            cache[key] = (
+                linv.t().to(device=device, dtype=dtype).contiguous(),
+                self.white_train(dtype=dtype, device=device),
+                float(-np.log(self.n) - 0.5 * self.log_det),
+            )
+        return cache[key]
+
+    def log_density_device(self, points: torch.Tensor) -> torch.Tensor:
+        """Device-resident log pdf for fp32 points already on the GPU."""
+        if self.prepare_failed:
+            return torch.full(
+                (points.shape[0],), float("-inf"), device=points.device
+            )
+        linv_t, xw, const = self.device_state(points.device, points.dtype)
+        y = (points @ linv_t).contiguous()
+        return ops.kde_logsumexp(y, xw) + const
+
     def log_density(self, points: torch.Tensor, device=None) -> torch.Tensor:
         """log pdf at [m, d] points.
 

@@ -357,6 +357,15 @@ class LSA(SA):
 
     def __call__(self, activations, predictions=None, num_threads=0):
         acts = _flatten_layers(activations)
+        if acts.is_cuda and self.kde is not None and not self.kde.prepare_failed:
+            # device fast path: column-select + whiten-GEMM + MFMA KDE kernel,
+            # no host round-trip
+            if self.removed_neurons:
+                keep = torch.from_numpy(
+                    np.delete(np.arange(acts.shape[1]), self.removed_neurons)
+                ).to(acts.device)
+                acts = acts.index_select(1, keep)
+            return -self.kde.log_density_device(acts.float().contiguous())
         dev = acts.device if acts.is_cuda else self.device
         acts = self._remove_unused_columns(acts.double().cpu())
         if self.kde is None:

@@ -22,6 +22,13 @@ _EXT = None
 _EXT_ERR = None
 
 
+def _load_compiled():
+    """Import the compiled _tip_hip extension module (raises ImportError)."""
+    from . import _tip_hip  # built in-tree by setup.py build_ext --inplace
+
+    return _tip_hip
+
+
 def _load_ext():
     """Load the HIP op wrapper (returns None if the .so is unavailable)."""
     global _EXT, _EXT_ERR

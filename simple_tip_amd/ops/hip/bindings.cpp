@@ -1,0 +1,255 @@
+// Python bindings for the simple_tip_amd HIP/CDNA4 kernels (gfx950).
+//
+// Native HIP throughout — no CUDA-compat paths; streams come from torch's
+// HIP stream pool so kernels serialize correctly with torch ops.
+
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+// launchers defined in the .hip translation units
+void launch_rownorm(const float*, int, int, float*, hipStream_t);
+void launch_pairwise_full(const float*, const float*, const float*,
+                          const float*, int, int, int, float*, hipStream_t);
+void launch_pairwise_rowmin(const float*, const float*, const float*,
+                            const float*, int, int, int, float*, int*, float*,
+                            int64_t*, hipStream_t);
+void launch_pairwise_kde(const float*, const float*, const float*,
+                         const float*, int, int, int, float2*, float*,
+                         hipStream_t);
+void launch_profile(int, const float*, const unsigned char*, const float*,
+                    const float*, float, int, int, int, int,
+                    unsigned long long*, long long*, hipStream_t);
+void launch_popcount(const unsigned long long*, int, int, long long*,
+                     hipStream_t);
+void launch_tknc(const float*, int, int, int, int, int, unsigned long long*,
+                 hipStream_t);
+void launch_bucketize(const double*, const double*, int, int, int,
+                      unsigned long long*, hipStream_t);
+void launch_cam_iteration(const unsigned long long*, int, int,
+                          unsigned long long*, unsigned char*, long long*,
+                          int*, long long*, hipStream_t);
+void launch_softmax_scores(const float*, int, int, float*, float*, float*,
+                           float*, hipStream_t);
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_f32_2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.dtype() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(t.dim() == 2 && t.is_contiguous(), name,
+              " must be 2D contiguous");
+}
+
+constexpr int kBN = 128;  // pairwise kernel column-block (matches pairwise.hip)
+
+torch::Tensor rownorm(torch::Tensor x) {
+  check_f32_2d(x, "x");
+  auto out = torch::empty({x.size(0)}, x.options());
+  launch_rownorm(x.data_ptr<float>(), x.size(0), x.size(1),
+                 out.data_ptr<float>(), cur_stream());
+  return out;
+}
+
+torch::Tensor pairwise_sqdist(torch::Tensor a, torch::Tensor b) {
+  check_f32_2d(a, "a");
+  check_f32_2d(b, "b");
+  TORCH_CHECK(a.size(1) == b.size(1), "feature dims differ");
+  const int m = a.size(0), n = b.size(0), k = a.size(1);
+  auto an = rownorm(a);
+  auto bn = rownorm(b);
+  auto out = torch::empty({m, n}, a.options());
+  launch_pairwise_full(a.data_ptr<float>(), b.data_ptr<float>(),
+                       an.data_ptr<float>(), bn.data_ptr<float>(), m, n, k,
+                       out.data_ptr<float>(), cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> rowmin_l2(torch::Tensor a, torch::Tensor b) {
+  check_f32_2d(a, "a");
+  check_f32_2d(b, "b");
+  TORCH_CHECK(a.size(1) == b.size(1), "feature dims differ");
+  TORCH_CHECK(b.size(0) > 0, "rowmin_l2 against empty set");
+  const int m = a.size(0), n = b.size(0), k = a.size(1);
+  const int jb = (n + kBN - 1) / kBN;
+  auto an = rownorm(a);
+  auto bn = rownorm(b);
+  auto pval = torch::empty({jb, m}, a.options());
+  auto pidx = torch::empty({jb, m}, a.options().dtype(torch::kInt32));
+  auto dist = torch::empty({m}, a.options());
+  auto idx = torch::empty({m}, a.options().dtype(torch::kInt64));
+  launch_pairwise_rowmin(a.data_ptr<float>(), b.data_ptr<float>(),
+                         an.data_ptr<float>(), bn.data_ptr<float>(), m, n, k,
+                         pval.data_ptr<float>(), pidx.data_ptr<int>(),
+                         dist.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                         cur_stream());
+  return {dist, idx};
+}
+
+torch::Tensor kde_logsumexp(torch::Tensor test, torch::Tensor train) {
+  check_f32_2d(test, "test");
+  check_f32_2d(train, "train");
+  TORCH_CHECK(test.size(1) == train.size(1), "feature dims differ");
+  TORCH_CHECK(train.size(0) > 0, "kde over empty train set");
+  const int m = test.size(0), n = train.size(0), k = test.size(1);
+  const int jb = (n + kBN - 1) / kBN;
+  auto an = rownorm(test);
+  auto bn = rownorm(train);
+  auto pkde = torch::empty({jb, m, 2}, test.options());
+  auto out = torch::empty({m}, test.options());
+  launch_pairwise_kde(test.data_ptr<float>(), train.data_ptr<float>(),
+                      an.data_ptr<float>(), bn.data_ptr<float>(), m, n, k,
+                      reinterpret_cast<float2*>(pkde.data_ptr<float>()),
+                      out.data_ptr<float>(), cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> profile(int mode, torch::Tensor acts,
+                                   torch::Tensor lo, torch::Tensor hi,
+                                   double thr, int64_t sections,
+                                   int64_t nbits) {
+  check_f32_2d(acts, "acts");
+  const int rows = acts.size(0), K = acts.size(1);
+  const int W = (nbits + 63) / 64;
+  auto words = torch::empty({rows, W}, acts.options().dtype(torch::kInt64));
+  auto scores = torch::empty({rows}, acts.options().dtype(torch::kInt64));
+  const float* lop = lo.defined() && lo.numel() ? lo.data_ptr<float>() : nullptr;
+  const float* hip = hi.defined() && hi.numel() ? hi.data_ptr<float>() : nullptr;
+  launch_profile(mode, acts.data_ptr<float>(), nullptr, lop, hip,
+                 static_cast<float>(thr), sections, rows, K, W,
+                 reinterpret_cast<unsigned long long*>(words.data_ptr<int64_t>()),
+                 reinterpret_cast<long long*>(scores.data_ptr<int64_t>()), cur_stream());
+  return {words, scores};
+}
+
+torch::Tensor pack_bits(torch::Tensor boolmat) {
+  TORCH_CHECK(boolmat.is_cuda() && boolmat.dtype() == torch::kBool &&
+              boolmat.dim() == 2 && boolmat.is_contiguous());
+  const int rows = boolmat.size(0), K = boolmat.size(1);
+  const int W = (K + 63) / 64;
+  auto words = torch::empty({rows, W},
+                            boolmat.options().dtype(torch::kInt64));
+  launch_profile(4 /*PROF_PACK*/, nullptr,
+                 boolmat.data_ptr<bool>()
+                     ? reinterpret_cast<const unsigned char*>(
+                           boolmat.data_ptr<bool>())
+                     : nullptr,
+                 nullptr, nullptr, 0.f, 1, rows, K, W,
+                 reinterpret_cast<unsigned long long*>(words.data_ptr<int64_t>()),
+                 nullptr, cur_stream());
+  return words;
+}
+
+torch::Tensor popcount_rows(torch::Tensor words) {
+  TORCH_CHECK(words.is_cuda() && words.dtype() == torch::kInt64 &&
+              words.dim() == 2 && words.is_contiguous());
+  auto out = torch::empty({words.size(0)}, words.options());
+  launch_popcount(
+      reinterpret_cast<const unsigned long long*>(words.data_ptr<int64_t>()),
+      words.size(0), words.size(1),
+      reinterpret_cast<long long*>(out.data_ptr<int64_t>()), cur_stream());
+  return out;
+}
+
+void tknc_layer(torch::Tensor layer, int64_t k, int64_t bit_offset,
+                torch::Tensor words) {
+  check_f32_2d(layer, "layer");
+  TORCH_CHECK(k >= 1 && k <= 4, "tknc supports k in [1,4]");
+  launch_tknc(layer.data_ptr<float>(), layer.size(0), layer.size(1), k,
+              bit_offset, words.size(1),
+              reinterpret_cast<unsigned long long*>(words.data_ptr<int64_t>()),
+              cur_stream());
+}
+
+torch::Tensor bucketize(torch::Tensor values, torch::Tensor thresholds,
+                        int64_t sections) {
+  TORCH_CHECK(values.is_cuda() && values.dtype() == torch::kFloat64);
+  TORCH_CHECK(thresholds.is_cuda() && thresholds.dtype() == torch::kFloat64);
+  const int n = values.size(0);
+  const int W = (sections + 63) / 64;
+  auto words = torch::empty({n, W}, values.options().dtype(torch::kInt64));
+  launch_bucketize(values.data_ptr<double>(), thresholds.data_ptr<double>(), n,
+                   sections, W,
+                   reinterpret_cast<unsigned long long*>(words.data_ptr<int64_t>()),
+                   cur_stream());
+  return words;
+}
+
+// Greedy CAM loop: device does all O(N*W) work, host only reads the picked
+// row per iteration. Returns the picked prefix (rows that added coverage).
+torch::Tensor cam_greedy(torch::Tensor words, int64_t nbits) {
+  TORCH_CHECK(words.is_cuda() && words.dtype() == torch::kInt64 &&
+              words.dim() == 2 && words.is_contiguous());
+  const int rows = words.size(0), W = words.size(1);
+  auto opts = words.options();
+  auto uncovered = torch::full({W}, -1, opts);  // all ones
+  const int tail = nbits % 64;
+  if (tail) uncovered.index_put_({W - 1}, (int64_t)((1ull << tail) - 1));
+  auto used = torch::zeros({rows}, opts.dtype(torch::kUInt8));
+  const int wpb = 8;
+  const int nblocks = (rows + wpb - 1) / wpb;
+  auto part_val = torch::empty({nblocks}, opts);
+  auto part_idx = torch::empty({nblocks}, opts.dtype(torch::kInt32));
+  auto result = torch::empty({2}, opts);
+  std::vector<int64_t> order;
+  order.reserve(std::min<int64_t>(rows, nbits));
+  auto stream = cur_stream();
+  long long host_result[2];
+  for (;;) {
+    launch_cam_iteration(
+        reinterpret_cast<const unsigned long long*>(words.data_ptr<int64_t>()),
+        rows, W,
+        reinterpret_cast<unsigned long long*>(uncovered.data_ptr<int64_t>()),
+        used.data_ptr<uint8_t>(),
+        reinterpret_cast<long long*>(part_val.data_ptr<int64_t>()),
+        part_idx.data_ptr<int>(),
+        reinterpret_cast<long long*>(result.data_ptr<int64_t>()), stream);
+    C10_HIP_CHECK(hipMemcpyAsync(host_result, result.data_ptr<int64_t>(),
+                                 2 * sizeof(long long), hipMemcpyDeviceToHost,
+                                 stream));
+    C10_HIP_CHECK(hipStreamSynchronize(stream));
+    if (host_result[0] < 0) break;
+    order.push_back(host_result[0]);
+    if ((int64_t)order.size() >= rows) break;
+  }
+  return torch::tensor(order, torch::dtype(torch::kInt64));
+}
+
+std::vector<torch::Tensor> softmax_scores(torch::Tensor probs) {
+  check_f32_2d(probs, "probs");
+  const int n = probs.size(0), c = probs.size(1);
+  auto neg_max = torch::empty({n}, probs.options());
+  auto neg_pcs = torch::empty({n}, probs.options());
+  auto entropy = torch::empty({n}, probs.options());
+  auto gini = torch::empty({n}, probs.options());
+  launch_softmax_scores(probs.data_ptr<float>(), n, c,
+                        neg_max.data_ptr<float>(), neg_pcs.data_ptr<float>(),
+                        entropy.data_ptr<float>(), gini.data_ptr<float>(),
+                        cur_stream());
+  return {neg_max, neg_pcs, entropy, gini};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "simple_tip_amd HIP/CDNA4 kernels (gfx950)";
+  m.def("rownorm", &rownorm);
+  m.def("pairwise_sqdist", &pairwise_sqdist);
+  m.def("rowmin_l2", &rowmin_l2);
+  m.def("kde_logsumexp", &kde_logsumexp);
+  m.def("profile", &profile);
+  m.def("pack_bits", &pack_bits);
+  m.def("popcount_rows", &popcount_rows);
+  m.def("tknc_layer", &tknc_layer);
+  m.def("bucketize", &bucketize);
+  m.def("cam_greedy", &cam_greedy);
+  m.def("softmax_scores", &softmax_scores);
+}

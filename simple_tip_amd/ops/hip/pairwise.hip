@@ -1,0 +1,279 @@
+// MFMA fp32 pairwise-distance kernel family for gfx950 (CDNA4).
+//
+// The workload's hot core: squared L2 distances
+// between a test-AT matrix A [M,K] and a train-AT matrix B [N,K], computed
+// as ||a||^2 + ||b||^2 - 2 a.b with the Gram matrix on the f32-input MFMA
+// (v_mfma_f32_32x32x2_f32: exact fp32 fmaf-chain numerics at the 157 TF f32
+// peak), staged through LDS in k-major tiles, with three fused epilogues:
+//   EPI_FULL   - write the D tile            (silhouette, debugging)
+//   EPI_ROWMIN - per-row min + argmin        (DSA, kmeans assign)
+//   EPI_KDE    - per-row online logsumexp(-d/2) partials (LSA KDE)
+// Per-(block-column) partials are combined by small deterministic
+// second-pass kernels so 1-GPU and sharded runs are bitwise reproducible.
+//
+// Geometry: 256 threads = 4 waves as 2x2; block tile 128x128, K-step 32;
+// each wave owns a 64x64 sub-tile = 2x2 MFMA 32x32 accumulators.
+
+#include "tip_common.h"
+
+#include <cfloat>
+#include <cmath>
+
+using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
+
+constexpr int BM = 128;
+constexpr int BN = 128;
+constexpr int BK = 32;
+constexpr int LDS_PAD = 4;  // pad k-major rows to de-conflict staging writes
+
+enum Epilogue { EPI_FULL = 0, EPI_ROWMIN = 1, EPI_KDE = 2 };
+
+// Stage a [BM rows x BK cols] block of src (row-major [rows,K]) into
+// k-major LDS tile lds[BK][BM+pad]: lds[k][r] = src[row0+r][k0+k].
+// 256 threads, 16 floats each as 4x float4 loads.
+TIP_DEV void stage_tile_kmajor(
+    float* lds, const float* __restrict__ src, int rows, int K, int row0, int k0) {
+  const int t = threadIdx.x;
+  const int r = t >> 1;                 // 0..127
+  const int kq = (t & 1) * (BK / 2);    // 0 or 16
+  const int grow = row0 + r;
+  const bool row_ok = grow < rows;
+#pragma unroll
+  for (int q4 = 0; q4 < BK / 2; q4 += 4) {
+    float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+    const int gk = k0 + kq + q4;
+    // float4 path only when rows are 16B-aligned (K % 4 == 0)
+    if (row_ok && gk + 3 < K && (K & 3) == 0) {
+      v = *reinterpret_cast<const float4*>(&src[(int64_t)grow * K + gk]);
+    } else if (row_ok) {
+      // K tail: scalar guarded loads (zeros contribute nothing to the dot)
+      float tmp[4] = {0.f, 0.f, 0.f, 0.f};
+      for (int e = 0; e < 4; ++e)
+        if (gk + e < K) tmp[e] = src[(int64_t)grow * K + gk + e];
+      v = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
+    }
+    lds[(kq + q4 + 0) * (BM + LDS_PAD) + r] = v.x;
+    lds[(kq + q4 + 1) * (BM + LDS_PAD) + r] = v.y;
+    lds[(kq + q4 + 2) * (BM + LDS_PAD) + r] = v.z;
+    lds[(kq + q4 + 3) * (BM + LDS_PAD) + r] = v.w;
+  }
+}
+
+template <int EPI>
+__launch_bounds__(256, 2) __global__ void pairwise_kernel(
+    const float* __restrict__ A,      // [M, K]
+    const float* __restrict__ B,      // [N, K]
+    const float* __restrict__ anorm,  // [M] row squared norms
+    const float* __restrict__ bnorm,  // [N]
+    int M, int N, int K,
+    float* __restrict__ out_full,     // EPI_FULL: [M, N]
+    float* __restrict__ pmin_val,     // EPI_ROWMIN: [jblocks, M]
+    int* __restrict__ pmin_idx,       //             [jblocks, M]
+    float2* __restrict__ pkde) {      // EPI_KDE:   [jblocks, M] (max, sum)
+  __shared__ float lds[2 * BK * (BM + LDS_PAD)];
+  float* As = lds;
+  float* Bs = lds + BK * (BM + LDS_PAD);
+  // reduction scratch: per block row x 2 column-halves
+  __shared__ float red_v[BM][2];
+  __shared__ int red_i[BM][2];
+  __shared__ float red_s[BM][2];
+
+  const int jblocks = gridDim.x;
+  const int bj = blockIdx.x;  // column block (inner, contiguous for L2)
+  const int bi = blockIdx.y;  // row block
+  const int row0 = bi * BM;
+  const int col0 = bj * BN;
+
+  const int lane = lane_id();
+  const int wid = wave_id();
+  const int wr = wid >> 1;  // wave row (0..1) -> 64 rows
+  const int wc = wid & 1;   // wave col (0..1) -> 64 cols
+
+  f32x16 acc[2][2] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    stage_tile_kmajor(As, A, M, K, row0, k0);
+    stage_tile_kmajor(Bs, B, N, K, col0, k0);
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const int k = kk + (lane >> 5);
+      const float* as = &As[k * (BM + LDS_PAD) + wr * 64 + (lane & 31)];
+      const float* bs = &Bs[k * (BM + LDS_PAD) + wc * 64 + (lane & 31)];
+      const float a0 = as[0], a1 = as[32];
+      const float b0 = bs[0], b1 = bs[32];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue ----
+  // C/D layout of v_mfma_f32_32x32x2_f32 (standard 32x32 map):
+  //   col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  const int jl0 = col0 + wc * 64 + (lane & 31);  // n=0 column
+  const float bn0 = (jl0 < N) ? bnorm[jl0] : 0.f;
+  const float bn1 = (jl0 + 32 < N) ? bnorm[jl0 + 32] : 0.f;
+
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+    // process the 16 regs; rows repeat across n, so handle n jointly
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int row_local = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+      const int block_row = wr * 64 + m * 32 + row_local;
+      const int i = row0 + block_row;
+      const float an = (i < M) ? anorm[i] : 0.f;
+      float d0 = an + bn0 - 2.f * acc[m][0][reg];
+      float d1 = an + bn1 - 2.f * acc[m][1][reg];
+      d0 = fmaxf(d0, 0.f);
+      d1 = fmaxf(d1, 0.f);
+      if (jl0 >= N) d0 = FLT_MAX;
+      if (jl0 + 32 >= N) d1 = FLT_MAX;
+
+      if (EPI == EPI_FULL) {
+        if (i < M) {
+          if (jl0 < N) out_full[(int64_t)i * N + jl0] = d0;
+          if (jl0 + 32 < N) out_full[(int64_t)i * N + jl0 + 32] = d1;
+        }
+      } else if (EPI == EPI_ROWMIN) {
+        MinIdx mi{d0, jl0};
+        mi = min_idx_combine(mi, MinIdx{d1, jl0 + 32});
+        mi = half_reduce_min(mi);
+        if ((lane & 31) == 0) {
+          red_v[block_row][wc] = mi.v;
+          red_i[block_row][wc] = mi.i;
+        }
+      } else {  // EPI_KDE: per-row max of t=-d/2 and sum exp(t - max)
+        MinIdx mi{d0, 0};
+        mi = min_idx_combine(mi, MinIdx{d1, 0});
+        mi = half_reduce_min(mi);
+        const float tmax = -0.5f * mi.v;  // = max_j(-d/2) over these 64 cols
+        float s = 0.f;
+        if (d0 != FLT_MAX) s += __expf(-0.5f * d0 - tmax);
+        if (d1 != FLT_MAX) s += __expf(-0.5f * d1 - tmax);
+        s = half_reduce_sum(s);
+        if ((lane & 31) == 0) {
+          red_v[block_row][wc] = tmax;
+          red_s[block_row][wc] = s;
+        }
+      }
+    }
+  }
+
+  if (EPI == EPI_FULL) return;
+  __syncthreads();
+  // combine the two column-halves (wc=0 covers lower j: ties prefer it)
+  for (int r = threadIdx.x; r < BM; r += blockDim.x) {
+    const int i = row0 + r;
+    if (i >= M) continue;
+    if (EPI == EPI_ROWMIN) {
+      MinIdx a{red_v[r][0], red_i[r][0]};
+      MinIdx b{red_v[r][1], red_i[r][1]};
+      MinIdx best = min_idx_combine(a, b);
+      pmin_val[(int64_t)bj * M + i] = best.v;
+      pmin_idx[(int64_t)bj * M + i] = best.i;
+    } else if (EPI == EPI_KDE) {
+      float m0 = red_v[r][0], s0 = red_s[r][0];
+      float m1 = red_v[r][1], s1 = red_s[r][1];
+      float mm, ss;
+      if (m0 >= m1) {
+        mm = m0;
+        ss = s0 + ((s1 > 0.f) ? s1 * __expf(m1 - m0) : 0.f);
+      } else {
+        mm = m1;
+        ss = s1 + ((s0 > 0.f) ? s0 * __expf(m0 - m1) : 0.f);
+      }
+      pkde[(int64_t)bj * M + i] = make_float2(mm, ss);
+    }
+  }
+}
+
+// Deterministic cross-block-column combines (ascending bj keeps np.argmin
+// lowest-index tie semantics; fixed order keeps results bitwise stable).
+__global__ void rowmin_combine_kernel(
+    const float* __restrict__ pval, const int* __restrict__ pidx, int jblocks,
+    int M, float* __restrict__ out_dist, int64_t* __restrict__ out_idx) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= M) return;
+  MinIdx best{FLT_MAX, 0x7fffffff};
+  for (int b = 0; b < jblocks; ++b) {
+    MinIdx cand{pval[(int64_t)b * M + i], pidx[(int64_t)b * M + i]};
+    best = min_idx_combine(best, cand);
+  }
+  out_dist[i] = sqrtf(best.v);
+  out_idx[i] = best.i;
+}
+
+__global__ void kde_combine_kernel(
+    const float2* __restrict__ pkde, int jblocks, int M,
+    float* __restrict__ out_lse) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= M) return;
+  float mm = -FLT_MAX, ss = 0.f;
+  for (int b = 0; b < jblocks; ++b) {
+    const float2 p = pkde[(int64_t)b * M + i];
+    if (p.y <= 0.f) continue;
+    if (p.x > mm) {
+      ss = p.y + ((ss > 0.f) ? ss * __expf(mm - p.x) : 0.f);
+      mm = p.x;
+    } else {
+      ss += p.y * __expf(p.x - mm);
+    }
+  }
+  out_lse[i] = (ss > 0.f) ? (mm + __logf(ss)) : -FLT_MAX;
+}
+
+// Row squared norms: one wave per row.
+__global__ void rownorm_kernel(
+    const float* __restrict__ X, int rows, int K, float* __restrict__ out) {
+  const int row = blockIdx.x * (blockDim.x / WAVE) + wave_id();
+  if (row >= rows) return;
+  const float* p = X + (int64_t)row * K;
+  float s = 0.f;
+  for (int k = lane_id(); k < K; k += WAVE) {
+    const float v = p[k];
+    s += v * v;
+  }
+  for (int off = 32; off >= 1; off >>= 1) s += __shfl_xor(s, off);
+  if (lane_id() == 0) out[row] = s;
+}
+
+// ---- host-side launchers (called from bindings.cpp) ----
+
+void launch_rownorm(const float* x, int rows, int k, float* out, hipStream_t s) {
+  const int wpb = 4;
+  rownorm_kernel<<<ceil_div(rows, wpb), wpb * WAVE, 0, s>>>(x, rows, k, out);
+}
+
+void launch_pairwise_full(const float* a, const float* b, const float* an,
+                          const float* bn, int m, int n, int k, float* out,
+                          hipStream_t s) {
+  dim3 grid(ceil_div(n, BN), ceil_div(m, BM));
+  pairwise_kernel<EPI_FULL><<<grid, 256, 0, s>>>(
+      a, b, an, bn, m, n, k, out, nullptr, nullptr, nullptr);
+}
+
+void launch_pairwise_rowmin(const float* a, const float* b, const float* an,
+                            const float* bn, int m, int n, int k,
+                            float* pval, int* pidx, float* out_dist,
+                            int64_t* out_idx, hipStream_t s) {
+  const int jb = ceil_div(n, BN);
+  dim3 grid(jb, ceil_div(m, BM));
+  pairwise_kernel<EPI_ROWMIN><<<grid, 256, 0, s>>>(
+      a, b, an, bn, m, n, k, nullptr, pval, pidx, nullptr);
+  rowmin_combine_kernel<<<ceil_div(m, 256), 256, 0, s>>>(
+      pval, pidx, jb, m, out_dist, out_idx);
+}
+
+void launch_pairwise_kde(const float* a, const float* b, const float* an,
+                         const float* bn, int m, int n, int k, float2* pkde,
+                         float* out_lse, hipStream_t s) {
+  const int jb = ceil_div(n, BN);
+  dim3 grid(jb, ceil_div(m, BM));
+  pairwise_kernel<EPI_KDE><<<grid, 256, 0, s>>>(
+      a, b, an, bn, m, n, k, nullptr, nullptr, nullptr, pkde);
+  kde_combine_kernel<<<ceil_div(m, 256), 256, 0, s>>>(pkde, jb, m, out_lse);
+}

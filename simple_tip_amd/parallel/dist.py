@@ -1,0 +1,96 @@
+"""torch.distributed helpers for data-parallel TIP sharding.
+
+The reference has NO inter-worker communication (SURVEY.md §2.4 — process
+pools sharing only the filesystem). Here test inputs shard across the 8
+GPUs of one MI355X node; per-input priority scores are all-gathered over
+RCCL/xGMI (backend "nccl" IS RCCL on ROCm). Messages are kilobytes — these
+collectives are latency-bound, the compute (pairwise kernels) dominates.
+Works identically with the gloo backend on CPU (world_size>1 unit tests).
+"""
+
+import datetime
+import os
+from typing import List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+
+def init_from_env(backend: Optional[str] = None) -> Tuple[int, int, torch.device]:
+    """Initialise from torchrun env vars; returns (rank, world, device).
+
+    Single-process (no WORLD_SIZE or WORLD_SIZE=1) leaves torch.distributed
+    uninitialised and returns rank 0.
+    """
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        dev = torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
+        return 0, 1, dev
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        dev = torch.device(f"cuda:{local_rank}")
+    else:
+        dev = torch.device("cpu")
+    if not dist.is_initialized():
+        dist.init_process_group(
+            backend=backend, timeout=datetime.timedelta(seconds=300)
+        )
+    return dist.get_rank(), dist.get_world_size(), dev
+
+
+def is_initialized() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def get_rank() -> int:
+    return dist.get_rank() if is_initialized() else 0
+
+
+def get_world_size() -> int:
+    return dist.get_world_size() if is_initialized() else 1
+
+
+def barrier():
+    if is_initialized():
+        dist.barrier()
+
+
+def shard_slice(n: int, rank: Optional[int] = None, world: Optional[int] = None) -> slice:
+    """Contiguous shard of [0, n) for this rank (first shards get the
+    remainder, matching torch.tensor_split)."""
+    rank = get_rank() if rank is None else rank
+    world = get_world_size() if world is None else world
+    base, rem = divmod(n, world)
+    start = rank * base + min(rank, rem)
+    return slice(start, start + base + (1 if rank < rem else 0))
+
+
+def allgather_rows(local: torch.Tensor, n_total: int) -> torch.Tensor:
+    """All-gather row shards produced by :func:`shard_slice` into the full
+    [n_total, ...] tensor (identical on every rank)."""
+    if not is_initialized():
+        return local
+    world = get_world_size()
+    pad = (n_total + world - 1) // world  # equal-size buffers (RCCL-safe)
+    buf = torch.zeros((pad,) + tuple(local.shape[1:]), dtype=local.dtype,
+                      device=local.device)
+    buf[: local.shape[0]] = local
+    shards = [torch.empty_like(buf) for _ in range(world)]
+    dist.all_gather(shards, buf)
+    parts = []
+    for r in range(world):
+        s = shard_slice(n_total, r, world)
+        parts.append(shards[r][: s.stop - s.start])
+    return torch.cat(parts, dim=0)
+
+
+def allreduce_max_scalar(value: float, device) -> float:
+    """MAX all-reduce of a python float (slowest-rank timing)."""
+    if not is_initialized():
+        return value
+    t = torch.tensor([value], dtype=torch.float64, device=device)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())
