@@ -67,6 +67,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--batch", type=int, default=10240, help="test inputs per GPU per step")
     ap.add_argument("--train-n", type=int, default=TRAIN_N)
+    ap.add_argument("--setup-epochs", type=int, default=2,
+                    help="untimed warm-up training epochs (class diversity)")
     args = ap.parse_args()
 
     rank, world, device = pdist.init_from_env()
@@ -76,19 +78,53 @@ def main():
 
     torch.manual_seed(0)  # identical weights on every rank
     model = ResNet20().to(device).eval()
+
+    # ---- setup (untimed) ----
+    # Brief training on class-structured synthetic data so the model's
+    # predictions spread over all classes: a random-init net predicts one
+    # class for everything, which would degenerate the per-class DSA/LSA
+    # work in the timed region (work-skipping = invalid measurement).
+    from simple_tip_amd.studies.synthetic import synthetic_images
+
+    tx, ty = synthetic_images("bench_cifar10", "warm", 10240, (3, 32, 32), 10)
+    tx_t = torch.from_numpy(tx)
+    ty_t = torch.from_numpy(ty)
+    opt = torch.optim.Adam(model.parameters(), lr=2e-3)
+    model.train()
+    for epoch in range(args.setup_epochs):
+        perm = torch.randperm(tx_t.shape[0])
+        for s in range(0, tx_t.shape[0], 512):
+            idx = perm[s : s + 512]
+            xb = tx_t[idx].to(device)
+            yb = ty_t[idx].to(device)
+            opt.zero_grad(set_to_none=True)
+            with torch.autocast("cuda", dtype=torch.bfloat16, enabled=on_gpu):
+                loss = torch.nn.functional.cross_entropy(model(xb).float(), yb)
+            loss.backward()
+            opt.step()
+    model.eval()
     if world > 1:
         # correctness by construction: rank-0 weights everywhere
         for p in model.parameters():
             torch.distributed.broadcast(p.data, src=0)
 
-    # ---- setup (untimed): train ATs + SA fits ----
+    # train-AT extraction + SA fits (the reference's "setup" bucket)
     g = torch.Generator().manual_seed(1234)
-    train_x = torch.randn(args.train_n, 3, 32, 32, generator=g)
+    base_x, base_y = synthetic_images(
+        "bench_cifar10", "train", args.train_n, (3, 32, 32), 10
+    )
+    train_x = torch.from_numpy(base_x)
     log(rank, f"extracting train ATs ({args.train_n} x 4096)")
     train_ats, train_probs = extract_ats(model, train_x, device)
     train_pred = train_probs.argmax(dim=1)
     del train_x, train_probs
+    if torch.unique(train_pred).numel() < 2:
+        # last-resort guard (should not trigger after warm-up training):
+        # use the synthetic labels so the per-class SA work is real
+        log(rank, "WARNING: predictions collapsed; using labels for SA fit")
+        train_pred = torch.from_numpy(base_y).to(train_pred.device)
 
+    log(rank, f"train pred classes: {torch.bincount(train_pred.cpu(), minlength=10).tolist()}")
     log(rank, "fitting DSA (0.3 subsample) + per-class LSA")
     dsa = DSA(train_ats, train_pred, subsampling=0.3, device=device)
     lsa = MultiModalSA.build_by_class(
@@ -97,13 +133,16 @@ def main():
     )
     train_classes = set(torch.unique(train_pred.cpu()).tolist())
 
-    # pre-generate per-rank test batches (distinct per step and rank)
-    n_pool = max(args.steps + args.warmup, 4)
-    gg = torch.Generator().manual_seed(10_000 + rank)
-    pool = [
-        torch.randn(args.batch, 3, 32, 32, generator=gg) for _ in range(min(n_pool, 8))
-    ]
-    labels = torch.randint(0, 10, (args.batch,), generator=gg)
+    # pre-generate per-rank test batches (distinct per step and rank, same
+    # class-structured distribution as training so predictions spread)
+    n_pool = min(max(args.steps + args.warmup, 4), 8)
+    pool, labels = [], None
+    for i in range(n_pool):
+        px, py = synthetic_images(
+            "bench_cifar10", f"test-r{rank}-{i}", args.batch, (3, 32, 32), 10
+        )
+        pool.append(torch.from_numpy(px))
+        labels = torch.from_numpy(py)
 
     def step(i):
         x = pool[i % len(pool)]
