@@ -60,6 +60,74 @@ def extract_ats(model, x, device, batch=1024):
     return torch.cat(ats), torch.cat(probs)
 
 
+class GraphedExtractor:
+    """Inference-path AT extractor: BN-folded bf16 channels_last model,
+    fixed-shape forward captured in a hipGraph (one replay per step).
+
+    Motivated by profiles/r01_bench_kernels.md: BN-inference launches and
+    small-conv launch overhead dominated the step before this.
+    """
+
+    def __init__(self, model, batch, device, use_graph=True):
+        from simple_tip_amd.models.fuse import fold_bn_inference
+
+        self.device = device
+        self.batch = batch
+        m = fold_bn_inference(model).to(device)
+        self.model = (
+            m.to(torch.bfloat16).to(memory_format=torch.channels_last)
+            if device.type == "cuda"
+            else m
+        )
+        self.graph = None
+        if use_graph and device.type == "cuda":
+            try:
+                self._capture()
+            except Exception as e:  # pragma: no cover - graph capture optional
+                print(f"[bench] hipGraph capture failed ({e!r}); eager path",
+                      file=sys.stderr)
+                self.graph = None
+
+    @torch.no_grad()
+    def _capture(self):
+        self.static_x = torch.empty(
+            self.batch, 3, 32, 32, device=self.device, dtype=torch.bfloat16
+        ).to(memory_format=torch.channels_last)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):  # warm up MIOpen finds before capture
+                self.model.forward_taps(self.static_x, AT_TAP)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            taps, logits = self.model.forward_taps(self.static_x, AT_TAP)
+            self.static_at = taps[0]
+            self.static_logits = logits
+        self.graph = g
+
+    @torch.no_grad()
+    def __call__(self, x):
+        if self.graph is not None and x.shape[0] == self.batch:
+            self.static_x.copy_(
+                x.to(self.device, torch.bfloat16, non_blocking=True).to(
+                    memory_format=torch.channels_last
+                )
+            )
+            self.graph.replay()
+            at = self.static_at
+            logits = self.static_logits
+        else:
+            xb = x.to(self.device)
+            if self.device.type == "cuda":
+                xb = xb.to(torch.bfloat16).to(memory_format=torch.channels_last)
+            taps, logits = self.model.forward_taps(xb, AT_TAP)
+            at = taps[0]
+        ats = at.reshape(at.shape[0], -1).float()
+        probs = torch.softmax(logits.float(), dim=1)
+        return ats, probs
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -114,10 +182,17 @@ def main():
         "bench_cifar10", "train", args.train_n, (3, 32, 32), 10
     )
     train_x = torch.from_numpy(base_x)
+    extractor = GraphedExtractor(model, args.batch, device, use_graph=on_gpu)
+
     log(rank, f"extracting train ATs ({args.train_n} x 4096)")
-    train_ats, train_probs = extract_ats(model, train_x, device)
-    train_pred = train_probs.argmax(dim=1)
-    del train_x, train_probs
+    at_parts, pred_parts = [], []
+    for s in range(0, train_x.shape[0], 1024):
+        a, p = extractor(train_x[s : s + 1024])
+        at_parts.append(a)
+        pred_parts.append(p.argmax(dim=1))
+    train_ats = torch.cat(at_parts)
+    train_pred = torch.cat(pred_parts)
+    del train_x, at_parts, pred_parts
     if torch.unique(train_pred).numel() < 2:
         # last-resort guard (should not trigger after warm-up training):
         # use the synthetic labels so the per-class SA work is real
@@ -146,7 +221,7 @@ def main():
 
     def step(i):
         x = pool[i % len(pool)]
-        ats, probs = extract_ats(model, x, device)
+        ats, probs = extractor(x)
         pred = probs.argmax(dim=1)
         # synthetic-data guard: random-init models may emit a class absent
         # from the train predictions; remap to a seen class

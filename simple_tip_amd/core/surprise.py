@@ -239,7 +239,16 @@ class MultiModalSA(SA):
         assert modal_idx.shape[0] == acts.shape[0], "discriminator length mismatch"
         if modal_idx.shape[0] == 0:
             return torch.empty(0)
-        res = torch.full((modal_idx.shape[0],), -float("inf"), dtype=torch.float64)
+        if acts.is_cuda:
+            # device-resident result: no per-mode host sync
+            res = torch.full(
+                (modal_idx.shape[0],), -float("inf"),
+                dtype=torch.float32, device=acts.device,
+            )
+        else:
+            res = torch.full(
+                (modal_idx.shape[0],), -float("inf"), dtype=torch.float64
+            )
         for modal_id in torch.unique(modal_idx).tolist():
             try:
                 sa = self.modal_sa[int(modal_id)]
@@ -251,7 +260,10 @@ class MultiModalSA(SA):
             a = acts[sel.to(acts.device)]
             p = None if preds is None else preds[sel.to(preds.device)]
             vals = sa(a, p)
-            res[sel] = vals.double().cpu()
+            if res.is_cuda:
+                res[sel.to(res.device)] = vals.float().to(res.device)
+            else:
+                res[sel] = vals.double().cpu()
         return res
 
 
@@ -328,6 +340,14 @@ class LSA(SA):
             warnings.warn(
                 "Feature removal dropped all ATs; this LSA instance will "
                 "always return density 0",
+                UserWarning,
+            )
+            return None
+        if cleaned.shape[0] < 2:
+            # scipy's gaussian_kde (and the reference) would crash outright
+            # on a singleton class; degrade to the prepare-failed behaviour
+            warnings.warn(
+                "LSA fit with < 2 samples; reporting density 0 for this mode",
                 UserWarning,
             )
             return None
