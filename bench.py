@@ -69,16 +69,18 @@ class GraphedExtractor:
     """
 
     def __init__(self, model, batch, device, use_graph=True):
-        from simple_tip_amd.models.fuse import fold_bn_inference
+        from simple_tip_amd.models.fuse import fold_bn_inference, pad_stem_channels
 
         self.device = device
         self.batch = batch
         m = fold_bn_inference(model).to(device)
-        self.model = (
-            m.to(torch.bfloat16).to(memory_format=torch.channels_last)
-            if device.type == "cuda"
-            else m
-        )
+        self.in_ch = 3
+        if device.type == "cuda":
+            pad_stem_channels(m, 4)  # NHWC bf16 igemm path needs >=4 channels
+            self.in_ch = 4
+            self.model = m.to(torch.bfloat16).to(memory_format=torch.channels_last)
+        else:
+            self.model = m
         self.graph = None
         if use_graph and device.type == "cuda":
             try:
@@ -90,8 +92,9 @@ class GraphedExtractor:
 
     @torch.no_grad()
     def _capture(self):
-        self.static_x = torch.empty(
-            self.batch, 3, 32, 32, device=self.device, dtype=torch.bfloat16
+        self.static_x = torch.zeros(
+            self.batch, self.in_ch, 32, 32, device=self.device,
+            dtype=torch.bfloat16,
         ).to(memory_format=torch.channels_last)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
@@ -106,22 +109,28 @@ class GraphedExtractor:
             self.static_logits = logits
         self.graph = g
 
+    def _prep(self, x):
+        xb = x.to(self.device, non_blocking=True)
+        if self.device.type == "cuda":
+            xb = xb.to(torch.bfloat16)
+            if self.in_ch > xb.shape[1]:
+                pad = torch.zeros(
+                    xb.shape[0], self.in_ch - xb.shape[1], *xb.shape[2:],
+                    dtype=xb.dtype, device=xb.device,
+                )
+                xb = torch.cat([xb, pad], dim=1)
+            xb = xb.to(memory_format=torch.channels_last)
+        return xb
+
     @torch.no_grad()
     def __call__(self, x):
         if self.graph is not None and x.shape[0] == self.batch:
-            self.static_x.copy_(
-                x.to(self.device, torch.bfloat16, non_blocking=True).to(
-                    memory_format=torch.channels_last
-                )
-            )
+            self.static_x.copy_(self._prep(x))
             self.graph.replay()
             at = self.static_at
             logits = self.static_logits
         else:
-            xb = x.to(self.device)
-            if self.device.type == "cuda":
-                xb = xb.to(torch.bfloat16).to(memory_format=torch.channels_last)
-            taps, logits = self.model.forward_taps(xb, AT_TAP)
+            taps, logits = self.model.forward_taps(self._prep(x), AT_TAP)
             at = taps[0]
         ats = at.reshape(at.shape[0], -1).float()
         probs = torch.softmax(logits.float(), dim=1)

@@ -466,6 +466,32 @@ class DSA(SA):
         self.train_predictions = preds
         self.num_classes = int(preds.max().item()) + 1
         self.badge_size = badge_size  # kept for API parity; kernels batch freely
+        self._class_cache = None  # device path: per-class (same_ats, b_table)
+
+    def _build_class_cache(self):
+        """Per class c: (same-class ATs contiguous, b_table) where
+        b_table[i] = distance from same-class AT i to its nearest
+        OTHER-class training AT.
+
+        dist_b depends only on which training AT is closest to the input
+        (two-hop), so it is a fixed function of the training set: we
+        precompute it once at fit time (part of the SA "setup" timing
+        bucket) and the per-input path becomes hop-1 + a gather.
+        """
+        cache = {}
+        for label in range(self.num_classes):
+            same_sel = self.train_predictions == label
+            same = self.train_activations[same_sel].contiguous()
+            if same.shape[0] == 0:
+                cache[label] = (None, None)
+                continue
+            other = self.train_activations[~same_sel].contiguous()
+            if other.shape[0] == 0:
+                cache[label] = (same, None)
+                continue
+            b_table, _ = ops.rowmin_l2(same, other)
+            cache[label] = (same, b_table)
+        self._class_cache = cache
 
     def __call__(self, activations, predictions, num_threads=None):
         target_ats = _flatten_layers(activations)
@@ -474,22 +500,29 @@ class DSA(SA):
         target_ats = target_ats.to(dev, self.train_activations.dtype)
         target_pred = target_pred.to(dev)
         dsa = torch.empty(target_pred.shape[0], dtype=target_ats.dtype, device=dev)
-        for label in torch.unique(target_pred).tolist():
+        if self._class_cache is None:
+            self._build_class_cache()
+        # single host sync: which classes are present and how many of each
+        counts = torch.bincount(
+            target_pred, minlength=self.num_classes
+        ).cpu().tolist()
+        for label in range(self.num_classes, len(counts)):
+            if counts[label]:
+                # predicted class never seen in training predictions:
+                # maximally surprising (the reference would crash here)
+                dsa[target_pred == label] = float("inf")
+        for label in range(self.num_classes):
+            if counts[label] == 0:
+                continue
             sel = target_pred == label
-            same = self.train_activations[self.train_predictions == label]
-            other = self.train_activations[self.train_predictions != label]
-            if same.shape[0] == 0:
-                # predicted class absent from the (subsampled) training
-                # predictions: maximally surprising. (The reference would
-                # crash here; only reachable with tiny/degenerate data.)
+            same, b_table = self._class_cache[label]
+            if same is None:
                 dsa[sel] = float("inf")
                 continue
             samples = target_ats[sel]
             dist_a, closest_idx = ops.rowmin_l2(samples, same)
-            if other.shape[0] == 0:
+            if b_table is None:
                 dsa[sel] = 0.0  # single-class training set: no contrast
                 continue
-            closest_ats = same[closest_idx]
-            dist_b, _ = ops.rowmin_l2(closest_ats, other)
-            dsa[sel] = dist_a / dist_b
+            dsa[sel] = dist_a / b_table[closest_idx]
         return dsa

@@ -14,6 +14,35 @@ from torch.nn.utils.fusion import fuse_conv_bn_eval
 from .cnn import ResNet20, _BasicBlock
 
 
+def pad_stem_channels(model, channels: int = 4):
+    """Zero-pad the first conv's input channels (3 -> 4) in place.
+
+    MIOpen has no NHWC bf16 igemm for 3-channel input and falls back to a
+    ~1.25 ms naive conv (profiles/r01_bench_kernels.md); a zero input
+    channel is mathematically identical and hits the fast path. Callers
+    must pad the input tensor to match.
+    """
+    import torch
+
+    first = None
+    for mod in model.modules():
+        if isinstance(mod, nn.Conv2d):
+            first = mod
+            break
+    assert first is not None and first.in_channels <= channels
+    if first.in_channels == channels:
+        return model
+    w = first.weight.data
+    neww = torch.zeros(
+        w.shape[0], channels, w.shape[2], w.shape[3],
+        dtype=w.dtype, device=w.device,
+    )
+    neww[:, : w.shape[1]] = w
+    first.weight = nn.Parameter(neww)
+    first.in_channels = channels
+    return model
+
+
 def fold_bn_inference(model: ResNet20) -> ResNet20:
     """Return an eval-mode deep copy with all conv+BN pairs fused.
 
