@@ -228,8 +228,18 @@ def main():
         pool.append(torch.from_numpy(px))
         labels = torch.from_numpy(py)
 
-    def step(i):
+    phase_log = os.environ.get("TIP_BENCH_PHASES") == "1"
+
+    def step(i, timed_phases=False):
+        marks = []
+
+        def mark(name):
+            if timed_phases:
+                torch.cuda.synchronize()
+                marks.append((name, time.perf_counter()))
+
         x = pool[i % len(pool)]
+        mark("start")
         ats, probs = extractor(x)
         pred = probs.argmax(dim=1)
         # synthetic-data guard: random-init models may emit a class absent
@@ -238,9 +248,13 @@ def main():
         ok = torch.isin(pred, seen)
         if not bool(ok.all()):
             pred = torch.where(ok, pred, seen[0])
+        mark("forward")
         unc = ops.softmax_uncertainties(probs)
+        mark("unc")
         dsa_scores = dsa(ats, pred)
+        mark("dsa")
         lsa_scores = lsa(ats, pred)
+        mark("lsa")
         gini = unc["deep_gini"]
         # publish score shards (tiny, latency-bound on xGMI)
         if world > 1:
@@ -248,11 +262,18 @@ def main():
             gini_all = pdist.allgather_rows(gini, n_total)
             dsa_all = pdist.allgather_rows(dsa_scores.float().to(device), n_total)
             _ = (gini_all, dsa_all)
+        mark("gather")
+        if timed_phases and marks:
+            parts = [
+                f"{marks[j][0]}={1000*(marks[j][1]-marks[j-1][1]):.1f}ms"
+                for j in range(1, len(marks))
+            ]
+            log(rank, "phases: " + " ".join(parts))
         return gini, dsa_scores, lsa_scores, pred
 
     log(rank, f"warmup x{args.warmup}")
     for i in range(args.warmup):
-        step(i)
+        step(i, timed_phases=phase_log and on_gpu and i == args.warmup - 1)
 
     pdist.barrier()
     if on_gpu:

@@ -485,12 +485,17 @@ class DSA(SA):
             if same.shape[0] == 0:
                 cache[label] = (None, None)
                 continue
+            same_norm = (
+                (same.float() * same.float()).sum(dim=1).contiguous()
+                if same.is_cuda
+                else None
+            )
             other = self.train_activations[~same_sel].contiguous()
             if other.shape[0] == 0:
-                cache[label] = (same, None)
+                cache[label] = (same, None, same_norm)
                 continue
             b_table, _ = ops.rowmin_l2(same, other)
-            cache[label] = (same, b_table)
+            cache[label] = (same, b_table, same_norm)
         self._class_cache = cache
 
     def __call__(self, activations, predictions, num_threads=None):
@@ -515,12 +520,12 @@ class DSA(SA):
             if counts[label] == 0:
                 continue
             sel = target_pred == label
-            same, b_table = self._class_cache[label]
+            same, b_table, same_norm = self._class_cache[label]
             if same is None:
                 dsa[sel] = float("inf")
                 continue
             samples = target_ats[sel]
-            dist_a, closest_idx = ops.rowmin_l2(samples, same)
+            dist_a, closest_idx = ops.rowmin_l2(samples, same, same_norm)
             if b_table is None:
                 dsa[sel] = 0.0  # single-class training set: no contrast
                 continue

@@ -94,8 +94,10 @@ def pairwise_sqdist(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return _route(a).pairwise_sqdist(a.contiguous(), b.contiguous())
 
 
-def rowmin_l2(a: torch.Tensor, b: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    return _route(a).rowmin_l2(a.contiguous(), b.contiguous())
+def rowmin_l2(
+    a: torch.Tensor, b: torch.Tensor, bnorm: torch.Tensor = None
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    return _route(a).rowmin_l2(a.contiguous(), b.contiguous(), bnorm)
 
 
 def kde_logsumexp(test_w: torch.Tensor, train_w: torch.Tensor) -> torch.Tensor:

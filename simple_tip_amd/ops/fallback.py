@@ -135,7 +135,7 @@ def pairwise_sqdist(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
 
 
 def rowmin_l2(
-    a: torch.Tensor, b: torch.Tensor, chunk: int = 4096
+    a: torch.Tensor, b: torch.Tensor, bnorm: torch.Tensor = None, chunk: int = 4096
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Per-row (min L2 distance, argmin) from rows of a to rows of b.
 

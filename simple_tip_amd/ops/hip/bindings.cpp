@@ -73,7 +73,8 @@ torch::Tensor pairwise_sqdist(torch::Tensor a, torch::Tensor b) {
   return out;
 }
 
-std::vector<torch::Tensor> rowmin_l2(torch::Tensor a, torch::Tensor b) {
+std::vector<torch::Tensor> rowmin_l2(torch::Tensor a, torch::Tensor b,
+                                     c10::optional<torch::Tensor> bnorm) {
   check_f32_2d(a, "a");
   check_f32_2d(b, "b");
   TORCH_CHECK(a.size(1) == b.size(1), "feature dims differ");
@@ -81,7 +82,7 @@ std::vector<torch::Tensor> rowmin_l2(torch::Tensor a, torch::Tensor b) {
   const int m = a.size(0), n = b.size(0), k = a.size(1);
   const int jb = (n + kBN - 1) / kBN;
   auto an = rownorm(a);
-  auto bn = rownorm(b);
+  auto bn = bnorm.has_value() ? bnorm.value() : rownorm(b);
   auto pval = torch::empty({jb, m}, a.options());
   auto pidx = torch::empty({jb, m}, a.options().dtype(torch::kInt32));
   auto dist = torch::empty({m}, a.options());
@@ -243,7 +244,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "simple_tip_amd HIP/CDNA4 kernels (gfx950)";
   m.def("rownorm", &rownorm);
   m.def("pairwise_sqdist", &pairwise_sqdist);
-  m.def("rowmin_l2", &rowmin_l2);
+  m.def("rowmin_l2", &rowmin_l2, py::arg("a"), py::arg("b"),
+        py::arg("bnorm") = py::none());
   m.def("kde_logsumexp", &kde_logsumexp);
   m.def("profile", &profile);
   m.def("pack_bits", &pack_bits);

@@ -53,8 +53,10 @@ def pairwise_sqdist(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return _ext.pairwise_sqdist(a.float(), b.float())
 
 
-def rowmin_l2(a: torch.Tensor, b: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    d, i = _ext.rowmin_l2(a.float(), b.float())
+def rowmin_l2(
+    a: torch.Tensor, b: torch.Tensor, bnorm: torch.Tensor = None
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    d, i = _ext.rowmin_l2(a.float(), b.float(), bnorm)
     return d, i
 
 
