@@ -224,6 +224,46 @@ torch::Tensor cam_greedy(torch::Tensor words, int64_t nbits) {
   return torch::tensor(order, torch::dtype(torch::kInt64));
 }
 
+// All-pairs Levenshtein distance matrix (CPU, threaded) for the text
+// corruptor's autocorrect dictionary (the reference uses the
+// polyleven pip package on a thread pool, text_corruptor.py:282-309).
+torch::Tensor levenshtein_matrix(std::vector<std::string> words) {
+  const int64_t n = (int64_t)words.size();
+  auto out = torch::zeros({n, n}, torch::dtype(torch::kUInt8));
+  auto acc = out.accessor<uint8_t, 2>();
+  at::parallel_for(0, n, 1, [&](int64_t begin, int64_t end) {
+    std::vector<int> dp0(64), dp1(64);
+    for (int64_t i = begin; i < end; ++i) {
+      const std::string& a = words[i];
+      const int la = (int)a.size();
+      for (int64_t j = i + 1; j < n; ++j) {
+        const std::string& b = words[j];
+        const int lb = (int)b.size();
+        if (lb + 1 > (int)dp0.size()) {
+          dp0.resize(lb + 1);
+          dp1.resize(lb + 1);
+        }
+        for (int c = 0; c <= lb; ++c) dp0[c] = c;
+        for (int r = 1; r <= la; ++r) {
+          dp1[0] = r;
+          const char ca = a[r - 1];
+          for (int c = 1; c <= lb; ++c) {
+            const int sub = dp0[c - 1] + (ca != b[c - 1] ? 1 : 0);
+            const int del = dp0[c] + 1;
+            const int ins = dp1[c - 1] + 1;
+            dp1[c] = std::min(sub, std::min(del, ins));
+          }
+          std::swap(dp0, dp1);
+        }
+        const int d = std::min(dp0[lb], 255);
+        acc[i][j] = (uint8_t)d;
+        acc[j][i] = (uint8_t)d;
+      }
+    }
+  });
+  return out;
+}
+
 std::vector<torch::Tensor> softmax_scores(torch::Tensor probs) {
   check_f32_2d(probs, "probs");
   const int n = probs.size(0), c = probs.size(1);
@@ -254,4 +294,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bucketize", &bucketize);
   m.def("cam_greedy", &cam_greedy);
   m.def("softmax_scores", &softmax_scores);
+  m.def("levenshtein_matrix", &levenshtein_matrix);
 }

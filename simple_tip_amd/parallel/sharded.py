@@ -1,0 +1,102 @@
+"""Train-AT-sharded reductions over RCCL/xGMI.
+
+The workload's large axis is the training-AT dimension of DSA/KDE
+(SURVEY.md §5 "long-context" analogue: up to 60k x 2304 floats). These
+helpers shard that axis across ranks: every rank holds a contiguous
+train-AT shard, scores the (replicated) test inputs against its shard with
+the local MFMA kernel, and the partials combine with small collectives —
+all_gather + deterministic rank-ordered merges, so 1-GPU and N-GPU results
+are bitwise identical.
+"""
+
+from typing import Tuple
+
+import torch
+import torch.distributed as dist
+
+from .. import ops
+from .dist import get_rank, get_world_size, is_initialized, shard_slice
+
+
+def shard_rows(full: torch.Tensor) -> Tuple[torch.Tensor, int]:
+    """This rank's contiguous row-shard of a replicated tensor and its
+    global row offset."""
+    s = shard_slice(full.shape[0])
+    return full[s].contiguous(), s.start
+
+
+def sharded_rowmin_l2(
+    test: torch.Tensor, train_shard: torch.Tensor, shard_offset: int
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Global (min L2 distance, global argmin) with the train rows sharded.
+
+    Partials merge in rank order with strict-less comparison, preserving the
+    lowest-global-index tie rule of the single-device path.
+    """
+    d, i = ops.rowmin_l2(test, train_shard)
+    i = i + shard_offset
+    if not is_initialized():
+        return d, i
+    world = get_world_size()
+    dg = [torch.empty_like(d) for _ in range(world)]
+    ig = [torch.empty_like(i) for _ in range(world)]
+    dist.all_gather(dg, d.contiguous())
+    dist.all_gather(ig, i.contiguous())
+    best_d, best_i = dg[0], ig[0]
+    for r in range(1, world):
+        # ranks hold ascending global offsets: strict less keeps lowest idx
+        take = dg[r] < best_d
+        best_d = torch.where(take, dg[r], best_d)
+        best_i = torch.where(take, ig[r], best_i)
+    return best_d, best_i
+
+
+def sharded_kde_logsumexp(
+    test_w: torch.Tensor, train_shard_w: torch.Tensor
+) -> torch.Tensor:
+    """Global logsumexp_i(-0.5 ||t - x_i||^2) with train rows sharded.
+
+    Per-rank partial LSEs merge with a rank-ordered streaming logsumexp —
+    the partial-reduction pattern this workload has in place of
+    ring-attention."""
+    part = ops.kde_logsumexp(test_w, train_shard_w)
+    if not is_initialized():
+        return part
+    world = get_world_size()
+    parts = [torch.empty_like(part) for _ in range(world)]
+    dist.all_gather(parts, part.contiguous())
+    stacked = torch.stack(parts)  # [world, m]
+    return torch.logsumexp(stacked, dim=0)
+
+
+def allreduce_minmax(mins: torch.Tensor, maxs: torch.Tensor):
+    """Cross-rank elementwise min/max of aggregate statistics (K18)."""
+    if is_initialized():
+        dist.all_reduce(mins, op=dist.ReduceOp.MIN)
+        dist.all_reduce(maxs, op=dist.ReduceOp.MAX)
+    return mins, maxs
+
+
+def allreduce_welford(count: float, mean: torch.Tensor, m2: torch.Tensor):
+    """Chan-merge Welford partials across ranks (deterministic rank order)."""
+    if not is_initialized():
+        return count, mean, m2
+    world = get_world_size()
+    dev = mean.device
+    counts = [torch.zeros(1, dtype=torch.float64, device=dev) for _ in range(world)]
+    means = [torch.empty_like(mean) for _ in range(world)]
+    m2s = [torch.empty_like(m2) for _ in range(world)]
+    dist.all_gather(counts, torch.tensor([count], dtype=torch.float64, device=dev))
+    dist.all_gather(means, mean.contiguous())
+    dist.all_gather(m2s, m2.contiguous())
+    tot_c, tot_mean, tot_m2 = float(counts[0].item()), means[0], m2s[0]
+    for r in range(1, world):
+        c = float(counts[r].item())
+        if c == 0:
+            continue
+        delta = means[r] - tot_mean
+        new_c = tot_c + c
+        tot_mean = tot_mean + delta * (c / new_c)
+        tot_m2 = tot_m2 + m2s[r] + delta * delta * (tot_c * c / new_c)
+        tot_c = new_c
+    return tot_c, tot_mean, tot_m2

@@ -47,7 +47,21 @@ def train_classifier(
         torch.manual_seed(seed)
     model = model.to(device)
     model.train()
-    opt = torch.optim.Adam(model.parameters(), lr=lr)
+
+    # Data-parallel training over RCCL/xGMI when torch.distributed is up
+    # (the 8-GPU active-learning retrain path). The models are <1M params,
+    # so DDP uses a single gradient bucket — one small all-reduce per step.
+    from ..parallel import dist as pdist
+
+    train_model = model
+    world = pdist.get_world_size()
+    if world > 1:
+        from torch.nn.parallel import DistributedDataParallel as DDP
+
+        ids = [device.index] if device.type == "cuda" else None
+        train_model = DDP(model, device_ids=ids, bucket_cap_mb=32)
+
+    opt = torch.optim.Adam(train_model.parameters(), lr=lr)
     loss_fn = nn.CrossEntropyLoss()
     n = x.shape[0]
     n_train = int(n * (1 - validation_split)) if validation_split else n
@@ -57,15 +71,24 @@ def train_classifier(
     yt = torch.as_tensor(np.asarray(y).reshape(-1)).long()
     use_amp = device.type == "cuda"
     for epoch in range(epochs):
-        perm = torch.randperm(n_train)
+        # identical permutation on every rank, rank-strided shard of it
+        gen = torch.Generator().manual_seed((seed or 0) * 1000 + epoch)
+        perm = torch.randperm(n_train, generator=gen)
+        rank = pdist.get_rank()
         total, correct, loss_sum = 0, 0, 0.0
-        for s in range(0, n_train, batch_size):
-            idx = perm[s : s + batch_size]
+        total_steps = max(1, (n_train + batch_size - 1) // batch_size)
+        # every rank runs the same number of steps (DDP all-reduce must
+        # match); overhanging ranks wrap to batch 0
+        for t in range((total_steps + world - 1) // world):
+            k = (t * world + rank) % total_steps
+            idx = perm[k * batch_size : (k + 1) * batch_size]
+            if idx.numel() == 0:
+                idx = perm[0:batch_size]
             xb = xt[idx].to(device, non_blocking=True)
             yb = yt[idx].to(device, non_blocking=True)
             opt.zero_grad(set_to_none=True)
             with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_amp):
-                logits = model(xb)
+                logits = train_model(xb)
                 loss = loss_fn(logits.float(), yb)
             loss.backward()
             opt.step()
