@@ -150,6 +150,10 @@ def main():
 
     rank, world, device = pdist.init_from_env()
     on_gpu = device.type == "cuda"
+    if on_gpu:
+        # exhaustive MIOpen find during (untimed) warmup picks the best conv
+        # algos for the fixed shapes before graph capture
+        torch.backends.cudnn.benchmark = True
     if on_gpu and not ops.hip_available():
         raise RuntimeError("bench on GPU requires the _tip_hip extension")
 

@@ -1,0 +1,85 @@
+"""GPU integration tests: the full engine on cuda:0 with the HIP ops
+engaged, cross-checked against the CPU path."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_coverage_worker_gpu_matches_cpu():
+    from simple_tip_amd.engine.coverage_handler import CoverageWorker
+    from simple_tip_amd.engine.model_handler import BaseModel
+    from simple_tip_amd.models import MnistCNN
+
+    torch.manual_seed(0)
+    model = MnistCNN().eval()
+    rng = np.random.RandomState(0)
+    train = rng.rand(96, 1, 28, 28).astype(np.float32)
+    test = rng.rand(64, 1, 28, 28).astype(np.float32)
+
+    cw_cpu = CoverageWorker(
+        BaseModel(model, [0, 1, 2, 3], device=torch.device("cpu"), predict_batch=32),
+        train,
+    )
+    t_cpu, s_cpu, o_cpu = cw_cpu.evaluate_all(test, "nominal")
+
+    model_gpu = MnistCNN().eval()
+    model_gpu.load_state_dict(model.state_dict())
+    model_gpu = model_gpu.cuda()
+    cw_gpu = CoverageWorker(
+        BaseModel(model_gpu, [0, 1, 2, 3], device=torch.device("cuda:0"), predict_batch=32),
+        train,
+    )
+    t_gpu, s_gpu, o_gpu = cw_gpu.evaluate_all(test, "nominal")
+
+    assert set(s_cpu) == set(s_gpu) and len(s_cpu) == 12
+    for metric in s_cpu:
+        # scores are integer popcounts; fp32-forward differences can only
+        # flip counts at exact threshold boundaries (measure-zero for NAC,
+        # rare for range-based metrics) — allow tiny drift
+        a, b = s_cpu[metric].astype(np.int64), s_gpu[metric].astype(np.int64)
+        frac = np.mean(a != b)
+        assert frac < 0.05, f"{metric}: {frac}"
+
+
+def test_surprise_handler_gpu_end_to_end():
+    from simple_tip_amd.engine.surprise_handler import SurpriseHandler
+    from simple_tip_amd.models import MnistCNN
+
+    torch.manual_seed(1)
+    model = MnistCNN().eval().cuda()
+    rng = np.random.RandomState(1)
+    train = rng.rand(400, 1, 28, 28).astype(np.float32)
+    nominal = rng.rand(96, 1, 28, 28).astype(np.float32)
+    ood = (rng.rand(96, 1, 28, 28) * 2).astype(np.float32)
+
+    sh = SurpriseHandler(model, sa_layers=[3], training_dataset=train,
+                         device=torch.device("cuda:0"), predict_batch=64)
+    res = sh.evaluate_all({"nominal": nominal, "ood": ood})
+    assert set(res) == {"dsa", "pc-lsa", "pc-mdsa", "pc-mlsa", "pc-mmdsa"}
+    for sa_name, per_ds in res.items():
+        for ds, (scores, cam_order, times) in per_ds.items():
+            assert scores.shape == (96,)
+            assert sorted(cam_order.tolist()) == list(range(96)), (sa_name, ds)
+            assert len(times) == 4
+    # metamorphic: scaled-up inputs are more surprising under dsa
+    assert np.nanmean(res["dsa"]["ood"][0]) > np.nanmean(res["dsa"]["nominal"][0])
+
+
+def test_prio_eval_gpu_artifacts(tmp_path, monkeypatch):
+    import simple_tip_amd.config as config
+    from simple_tip_amd.studies import get_case_study
+
+    monkeypatch.setattr(config, "OUTPUT_FOLDER", str(tmp_path))
+    study = get_case_study("mnist", scale=0.003, device=torch.device("cuda:0"))
+    study.train([0])
+    study.run_prio_eval([0])
+    prio = tmp_path / "priorities"
+    files = {f.name for f in prio.iterdir()}
+    assert "mnist_nominal_0_is_misclassified.npy" in files
+    assert "mnist_nominal_0_dsa_scores.npy" in files
+    assert "mnist_ood_0_KMNC_2_cam_order.npy" in files
+    dsa = np.load(prio / "mnist_nominal_0_dsa_scores.npy")
+    assert np.isfinite(dsa).all() and (dsa >= 0).all()
