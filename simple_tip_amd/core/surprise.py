@@ -483,7 +483,7 @@ class DSA(SA):
             same_sel = self.train_predictions == label
             same = self.train_activations[same_sel].contiguous()
             if same.shape[0] == 0:
-                cache[label] = (None, None)
+                cache[label] = (None, None, None)
                 continue
             same_norm = (
                 (same.float() * same.float()).sum(dim=1).contiguous()

@@ -28,20 +28,29 @@ constexpr int LDS_PAD = 4;  // pad k-major rows to de-conflict staging writes
 
 enum Epilogue { EPI_FULL = 0, EPI_ROWMIN = 1, EPI_KDE = 2 };
 
-// Stage a [BM rows x BK cols] block of src (row-major [rows,K]) into
-// k-major LDS tile lds[BK][BM+pad]: lds[k][r] = src[row0+r][k0+k].
-// 256 threads, 16 floats each as 4x float4 loads.
-TIP_DEV void stage_tile_kmajor(
-    float* lds, const float* __restrict__ src, int rows, int K, int row0, int k0) {
+// Staging is split T14-style (issue-early / write-late): the global loads
+// for tile t+1 are issued into registers BEFORE tile t's MFMA loop (HBM
+// latency hides under the ~4k-cycle f32-MFMA compute phase) and the LDS
+// write pass runs after the barrier. LDS image is k-major
+// lds[BK][BM+pad]: lds[k][r] = src[row0+r][k0+k]; 256 threads, 16 floats
+// each (4x float4).
+
+struct StageRegs {
+  float4 v[4];
+};
+
+TIP_DEV StageRegs stage_load(
+    const float* __restrict__ src, int rows, int K, int row0, int k0) {
+  StageRegs sr;
   const int t = threadIdx.x;
   const int r = t >> 1;                 // 0..127
   const int kq = (t & 1) * (BK / 2);    // 0 or 16
   const int grow = row0 + r;
   const bool row_ok = grow < rows;
 #pragma unroll
-  for (int q4 = 0; q4 < BK / 2; q4 += 4) {
+  for (int q4 = 0; q4 < 4; ++q4) {
     float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
-    const int gk = k0 + kq + q4;
+    const int gk = k0 + kq + q4 * 4;
     // float4 path only when rows are 16B-aligned (K % 4 == 0)
     if (row_ok && gk + 3 < K && (K & 3) == 0) {
       v = *reinterpret_cast<const float4*>(&src[(int64_t)grow * K + gk]);
@@ -52,10 +61,21 @@ TIP_DEV void stage_tile_kmajor(
         if (gk + e < K) tmp[e] = src[(int64_t)grow * K + gk + e];
       v = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
     }
-    lds[(kq + q4 + 0) * (BM + LDS_PAD) + r] = v.x;
-    lds[(kq + q4 + 1) * (BM + LDS_PAD) + r] = v.y;
-    lds[(kq + q4 + 2) * (BM + LDS_PAD) + r] = v.z;
-    lds[(kq + q4 + 3) * (BM + LDS_PAD) + r] = v.w;
+    sr.v[q4] = v;
+  }
+  return sr;
+}
+
+TIP_DEV void stage_write(float* lds, const StageRegs& sr) {
+  const int t = threadIdx.x;
+  const int r = t >> 1;
+  const int kq = (t & 1) * (BK / 2);
+#pragma unroll
+  for (int q4 = 0; q4 < 4; ++q4) {
+    lds[(kq + q4 * 4 + 0) * (BM + LDS_PAD) + r] = sr.v[q4].x;
+    lds[(kq + q4 * 4 + 1) * (BM + LDS_PAD) + r] = sr.v[q4].y;
+    lds[(kq + q4 * 4 + 2) * (BM + LDS_PAD) + r] = sr.v[q4].z;
+    lds[(kq + q4 * 4 + 3) * (BM + LDS_PAD) + r] = sr.v[q4].w;
   }
 }
 
@@ -78,9 +98,20 @@ __launch_bounds__(256, 2) __global__ void pairwise_kernel(
   __shared__ int red_i[BM][2];
   __shared__ float red_s[BM][2];
 
+  // XCD-aware block swizzle (T1, bijective form): consecutive remapped ids
+  // land on one XCD and share the A panel in that XCD's L2.
   const int jblocks = gridDim.x;
-  const int bj = blockIdx.x;  // column block (inner, contiguous for L2)
-  const int bi = blockIdx.y;  // row block
+  int bi, bj;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    const int id = blockIdx.y * gridDim.x + blockIdx.x;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = id % 8, pos = id / 8;
+    const int newid =
+        (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+    bi = newid / gridDim.x;
+    bj = newid % gridDim.x;
+  }
   const int row0 = bi * BM;
   const int col0 = bj * BN;
 
@@ -91,10 +122,20 @@ __launch_bounds__(256, 2) __global__ void pairwise_kernel(
 
   f32x16 acc[2][2] = {};
 
+  // T14 pipeline: prologue stages tile 0; each iteration issues tile t+1's
+  // global loads before tile t's MFMAs and writes them to LDS afterwards.
+  StageRegs ra = stage_load(A, M, K, row0, 0);
+  StageRegs rb = stage_load(B, N, K, col0, 0);
+  stage_write(As, ra);
+  stage_write(Bs, rb);
+  __syncthreads();
+
   for (int k0 = 0; k0 < K; k0 += BK) {
-    stage_tile_kmajor(As, A, M, K, row0, k0);
-    stage_tile_kmajor(Bs, B, N, K, col0, k0);
-    __syncthreads();
+    const bool has_next = (k0 + BK) < K;
+    if (has_next) {
+      ra = stage_load(A, M, K, row0, k0 + BK);
+      rb = stage_load(B, N, K, col0, k0 + BK);
+    }
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 2) {
       const int k = kk + (lane >> 5);
@@ -107,7 +148,12 @@ __launch_bounds__(256, 2) __global__ void pairwise_kernel(
       acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
       acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
     }
-    __syncthreads();
+    __syncthreads();  // all reads of tile t done
+    if (has_next) {
+      stage_write(As, ra);
+      stage_write(Bs, rb);
+    }
+    __syncthreads();  // tile t+1 visible
   }
 
   // ---- epilogue ----
