@@ -5,6 +5,8 @@ batch 32, predict batch 600, dsa_badge_size 500, num_selected 2500,
 SA [5], NC [3, 5] — the int-valued tap entries; the reference's tuple
 entries are dead config)."""
 
+import numpy as np
+
 from ..config import StudyConfig
 from ..models.transformer import ImdbTransformer
 from .base import CaseStudy
@@ -34,6 +36,57 @@ class ImdbCaseStudy(CaseStudy):
 
     def build_model(self):
         return ImdbTransformer(vocab_size=VOCAB_SIZE, maxlen=INPUT_MAXLEN)
+
+    def load_text_datasets(self, n_train=None, n_test=None, severity=0.5, seed=0):
+        """Full reference-shaped text pipeline: synthetic texts ->
+        TextCorruptor (IMDB-C, severity 0.5, seed 0 — reference
+        case_study_imdb.py:316-319) -> Tokenizer(2000) -> pad to maxlen
+        (case_study_imdb.py:321-336). Returns token arrays like
+        :meth:`load_datasets`.
+        """
+        from ..core.text_corruptor import TextCorruptor
+        from ..utils.tokenizer import Tokenizer, pad_sequences
+        from .synthetic import _rng, make_ood_split
+
+        cfg = self.config
+        n_train = n_train or self._n(cfg.train_size)
+        n_test = n_test or self._n(cfg.test_size)
+
+        # deterministic pseudo-word vocabulary with class-dependent usage
+        vrng = _rng(cfg.name, "text-vocab")
+        letters = "abcdefghijklmnopqrstuvwxyz"
+        vocab = sorted(
+            {
+                "".join(vrng.choice(list(letters), size=vrng.randint(5, 9)))
+                for _ in range(3000)
+            }
+        )
+        boost = _rng(cfg.name, "text-classes").rand(cfg.num_classes, len(vocab)) ** 4
+        base = 1.0 / (np.arange(len(vocab)) + 10.0)
+
+        def make_texts(split, n):
+            rng = _rng(cfg.name, split)
+            y = rng.randint(0, cfg.num_classes, size=n)
+            texts = []
+            for i in range(n):
+                p = base * (1.0 + 8.0 * boost[y[i]])
+                p /= p.sum()
+                words = rng.choice(vocab, size=INPUT_MAXLEN + 20, p=p)
+                texts.append(" ".join(words))
+            return texts, y.astype(np.int64)
+
+        train_texts, train_y = make_texts("text-train", n_train)
+        nom_texts, nom_y = make_texts("text-test", n_test)
+        cor_src, cor_y = make_texts("text-corrupt-src", n_test)
+        corruptor = TextCorruptor(train_texts, dict_size=2000)
+        cor_texts = corruptor.corrupt(cor_src, severity=severity, seed=seed)
+
+        tok = Tokenizer(num_words=VOCAB_SIZE).fit_on_texts(train_texts)
+        to_arr = lambda ts: pad_sequences(tok.texts_to_sequences(ts), INPUT_MAXLEN)
+        train = (to_arr(train_texts), train_y)
+        nominal = (to_arr(nom_texts), nom_y)
+        ood = make_ood_split(nominal[0], nominal[1], to_arr(cor_texts), cor_y)
+        return train, nominal, ood
 
     def load_datasets(self):
         cfg = self.config
