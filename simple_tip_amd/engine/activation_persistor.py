@@ -9,7 +9,7 @@ from typing import Dict, List
 
 import numpy as np
 
-from ..config import OUTPUT_FOLDER
+from .. import config
 from .model_handler import BaseModel, iter_batches
 
 BADGE_SIZE = 100
@@ -36,7 +36,7 @@ def persist(
         ):
             for i, layer in enumerate(acts):
                 folder = os.path.join(
-                    OUTPUT_FOLDER,
+                    config.OUTPUT_FOLDER,
                     "activations",
                     case_study,
                     f"model_{model_id}",

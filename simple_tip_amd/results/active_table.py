@@ -14,7 +14,8 @@ from typing import Dict, List, Tuple
 import numpy as np
 import pandas as pd
 
-from ..config import OUTPUT_FOLDER, ensure_dir
+from .. import config
+from ..config import ensure_dir
 from .common import APPROACHES, CASE_STUDIES, category
 
 logger = logging.getLogger(__name__)
@@ -24,7 +25,7 @@ RANDOM = "random"
 
 def load_runs(case_study: str) -> Dict[str, Dict[str, Dict[int, dict]]]:
     """{approach: {ood_or_nom: {model_id: {(split, obs/fut): acc}}}}."""
-    folder = os.path.join(OUTPUT_FOLDER, "active_learning")
+    folder = os.path.join(config.OUTPUT_FOLDER, "active_learning")
     res: Dict[str, Dict[str, Dict[int, dict]]] = {}
     if not os.path.isdir(folder):
         return res
@@ -91,6 +92,6 @@ def run(case_studies=None) -> pd.DataFrame:
     """Generate results/active.csv."""
     df = build_dataframe(case_studies)
     ensure_dir("results")
-    df.to_csv(os.path.join(OUTPUT_FOLDER, "results", "active.csv"))
-    logger.info("wrote %s/results/active.csv", OUTPUT_FOLDER)
+    df.to_csv(os.path.join(config.OUTPUT_FOLDER, "results", "active.csv"))
+    logger.info("wrote %s/results/active.csv", config.OUTPUT_FOLDER)
     return df

@@ -15,7 +15,8 @@ from typing import Dict
 import numpy as np
 import pandas as pd
 
-from ..config import OUTPUT_FOLDER, ensure_dir
+from .. import config
+from ..config import ensure_dir
 from ..core.apfd import apfd_from_order
 from .common import (
     APPROACHES,
@@ -72,7 +73,7 @@ def load_apfd_values(case_study: str, ds_name: str) -> Dict[str, Dict[int, float
 def _load_times(case_studies) -> Dict[tuple, list]:
     """{(cs, ds, model_id, metric): [setup, pred, quant(, cam)]} pickles,
     first 10 models only."""
-    folder = os.path.join(OUTPUT_FOLDER, "times")
+    folder = os.path.join(config.OUTPUT_FOLDER, "times")
     res = {}
     if not os.path.isdir(folder):
         return res
@@ -137,16 +138,16 @@ def run(case_studies=None) -> pd.DataFrame:
     """Generate results/apfds.csv (+ LaTeX paper excerpt)."""
     df = build_dataframe(case_studies)
     ensure_dir("results")
-    df.to_csv(os.path.join(OUTPUT_FOLDER, "results", "apfds.csv"))
+    df.to_csv(os.path.join(config.OUTPUT_FOLDER, "results", "apfds.csv"))
     try:
         paper = df.iloc[
             df.index.get_level_values("approach").isin(PAPER_APPROACHES)
         ]
         with open(
-            os.path.join(OUTPUT_FOLDER, "results", "apfd_paper_table.tex"), "w"
+            os.path.join(config.OUTPUT_FOLDER, "results", "apfd_paper_table.tex"), "w"
         ) as f:
             f.write(paper.to_latex(multicolumn_format="c", multirow=True))
     except Exception as e:  # noqa: BLE001 - latex formatting is best-effort
         warnings.warn(f"LaTeX table generation failed: {e}")
-    logger.info("wrote %s/results/apfds.csv", OUTPUT_FOLDER)
+    logger.info("wrote %s/results/apfds.csv", config.OUTPUT_FOLDER)
     return df

@@ -7,7 +7,7 @@ from typing import Dict, List, Tuple
 
 import numpy as np
 
-from ..config import OUTPUT_FOLDER
+from .. import config
 
 NUM_RUNS = 100
 N_FIRST_MODELS_FOR_TIMES = 10
@@ -56,7 +56,7 @@ def parse_priority_filename(
 
 def iter_priority_files(case_study: str, ds_name: str):
     """Yield (model_id, data_type, path) for every matching artifact."""
-    folder = os.path.join(OUTPUT_FOLDER, "priorities")
+    folder = os.path.join(config.OUTPUT_FOLDER, "priorities")
     if not os.path.isdir(folder):
         return
     prefix = f"{case_study}_{ds_name}_"

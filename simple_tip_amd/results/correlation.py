@@ -18,7 +18,8 @@ import numpy as np
 import pandas as pd
 from scipy.stats import wilcoxon
 
-from ..config import OUTPUT_FOLDER, ensure_dir
+from .. import config
+from ..config import ensure_dir
 from .active_table import load_runs
 from .apfd_table import load_apfd_values
 from .common import APPROACHES, CASE_STUDIES
@@ -67,10 +68,10 @@ def _pairwise(measurements: Dict[str, Dict[str, float]], approaches: List[str]):
 def _write(exp: str, approaches, p, e):
     ensure_dir("results")
     pd.DataFrame(p, index=approaches, columns=approaches).to_csv(
-        os.path.join(OUTPUT_FOLDER, "results", f"{exp}_correlation_p.csv")
+        os.path.join(config.OUTPUT_FOLDER, "results", f"{exp}_correlation_p.csv")
     )
     pd.DataFrame(e, index=approaches, columns=approaches).to_csv(
-        os.path.join(OUTPUT_FOLDER, "results", f"{exp}_correlation_eff.csv")
+        os.path.join(config.OUTPUT_FOLDER, "results", f"{exp}_correlation_eff.csv")
     )
     try:
         import matplotlib  # noqa: F401
@@ -95,7 +96,7 @@ def _plot_heatmap(exp, approaches, p, e):  # pragma: no cover - optional dep
     ax.set_yticklabels(approaches)
     fig.colorbar(im)
     fig.tight_layout()
-    fig.savefig(os.path.join(OUTPUT_FOLDER, "results", f"{exp}_correlation.png"))
+    fig.savefig(os.path.join(config.OUTPUT_FOLDER, "results", f"{exp}_correlation.png"))
     plt.close(fig)
 
 
