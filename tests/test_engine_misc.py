@@ -65,7 +65,6 @@ def test_activation_persistor_layout(tmp_path, monkeypatch):
     import simple_tip_amd.engine.activation_persistor as ap
 
     monkeypatch.setattr(config, "OUTPUT_FOLDER", str(tmp_path))
-    monkeypatch.setattr(ap, "OUTPUT_FOLDER", str(tmp_path))
     from simple_tip_amd.models import MnistCNN
 
     model = MnistCNN().eval()
