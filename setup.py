@@ -27,6 +27,7 @@ setup(
                 os.path.join(HIP_DIR, "pairwise.hip"),
                 os.path.join(HIP_DIR, "coverage.hip"),
                 os.path.join(HIP_DIR, "cam.hip"),
+                os.path.join(HIP_DIR, "resnet_fused.hip"),
                 os.path.join(HIP_DIR, "scores.hip"),
             ],
             extra_compile_args={

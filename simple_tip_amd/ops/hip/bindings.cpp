@@ -35,6 +35,14 @@ void launch_cam_iteration(const unsigned long long*, int, int,
                           int*, long long*, hipStream_t);
 void launch_softmax_scores(const float*, int, int, float*, float*, float*,
                            float*, hipStream_t);
+void launch_mfma_probe(const short*, const short*, float*, hipStream_t);
+void launch_resblock(int, int, const short*, short*, const short*,
+                     const float*, const short*, const float*, hipStream_t);
+void launch_downblock(int, int, const short*, short*, const short*,
+                      const float*, const short*, const float*, const short*,
+                      const float*, hipStream_t);
+void launch_stem(int, const short*, short*, const short*, const float*,
+                 hipStream_t);
 
 namespace {
 
@@ -278,6 +286,53 @@ std::vector<torch::Tensor> softmax_scores(torch::Tensor probs) {
   return {neg_max, neg_pcs, entropy, gini};
 }
 
+// ---- fused ResNet-20 inference (bf16 NHWC) ----
+
+const short* bf16_ptr(const torch::Tensor& t) {
+  TORCH_CHECK(t.is_cuda() && t.dtype() == torch::kBFloat16 && t.is_contiguous());
+  return reinterpret_cast<const short*>(t.data_ptr());
+}
+
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
+  auto d = torch::empty({16, 16}, a.options().dtype(torch::kFloat32));
+  launch_mfma_probe(bf16_ptr(a), bf16_ptr(b),
+                    d.data_ptr<float>(), cur_stream());
+  return d;
+}
+
+torch::Tensor resnet_block(int64_t variant, torch::Tensor x, torch::Tensor w1,
+                           torch::Tensor b1, torch::Tensor w2, torch::Tensor b2) {
+  const int batch = x.size(0);
+  auto out = torch::empty_like(x);
+  launch_resblock(variant, batch, bf16_ptr(x),
+                  const_cast<short*>(bf16_ptr(out)), bf16_ptr(w1),
+                  b1.data_ptr<float>(), bf16_ptr(w2), b2.data_ptr<float>(),
+                  cur_stream());
+  return out;
+}
+
+torch::Tensor resnet_down(int64_t variant, torch::Tensor x, torch::Tensor w1,
+                          torch::Tensor b1, torch::Tensor w2, torch::Tensor b2,
+                          torch::Tensor wsc, torch::Tensor bsc) {
+  const int batch = x.size(0);
+  const int64_t out_elems = x.size(1) / 2;  // H*W*C -> (H/2)(W/2)(2C)
+  auto out = torch::empty({batch, out_elems}, x.options());
+  launch_downblock(variant, batch, bf16_ptr(x),
+                   const_cast<short*>(bf16_ptr(out)), bf16_ptr(w1),
+                   b1.data_ptr<float>(), bf16_ptr(w2), b2.data_ptr<float>(),
+                   bf16_ptr(wsc), bsc.data_ptr<float>(), cur_stream());
+  return out;
+}
+
+torch::Tensor resnet_stem(torch::Tensor x, torch::Tensor w, torch::Tensor b) {
+  const int batch = x.size(0);
+  const int64_t out_elems = 32 * 32 * 16;
+  auto out = torch::empty({batch, out_elems}, x.options());
+  launch_stem(batch, bf16_ptr(x), const_cast<short*>(bf16_ptr(out)),
+              bf16_ptr(w), b.data_ptr<float>(), cur_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -295,4 +350,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cam_greedy", &cam_greedy);
   m.def("softmax_scores", &softmax_scores);
   m.def("levenshtein_matrix", &levenshtein_matrix);
+  m.def("mfma_probe", &mfma_probe);
+  m.def("resnet_block", &resnet_block);
+  m.def("resnet_down", &resnet_down);
+  m.def("resnet_stem", &resnet_stem);
 }

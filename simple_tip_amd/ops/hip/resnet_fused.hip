@@ -1,0 +1,354 @@
+// Fused ResNet-20 inference kernels for gfx950 (bf16 NHWC, CDNA4 MFMA).
+//
+// Motivation (profiles/r01_bench_kernels.md): MIOpen runs the bench's
+// AT-extraction forward at ~26 TF effective — per-layer launches, im2col
+// staging and BatchNorm inference dominate. These kernels compute one
+// RESIDUAL BLOCK per launch: the input plane (with 1-pixel halo) is staged
+// into LDS once, conv1 -> relu writes the intermediate plane to LDS only,
+// conv2 + bias + residual-add + relu writes the block output to HBM. HBM
+// traffic per block = read input + write output; the intermediate never
+// leaves the CU.
+//
+// Layout/geometry:
+// - activations: NHWC bf16; LDS image [(H+2) x (W+2) x C] with zeroed halo,
+//   16-byte units XOR-swizzled (u ^= (u>>4)&7) so the per-tap
+//   ds_read_b128 gathers are bank-conflict-free for C in {16, 32, 64}.
+// - MFMA: v_mfma_f32_16x16x32_bf16. M = 16 consecutive output pixels
+//   (row-major in the plane), N = 16 output channels, K = 32 consecutive
+//   (tap, channel) pairs of the 3x3xC patch (zero-padded to a multiple
+//   of 32). A-fragment: lane l holds A[i = l&15][k = (l>>4)*8 + e]
+//   (one 16-B LDS read per lane); B-fragment comes pre-packed on the host
+//   into [ksteps][64][8] bf16 so each lane reads its 16 B straight from
+//   global (L2-resident; weights are <= 73 KB per layer).
+//   C/D: col = lane&15, row = (lane>>4)*4 + reg.
+//   (The layout is verified at runtime by the mfma_probe test kernel.)
+// - one workgroup = one image; 4 waves split the plane's 16-pixel tiles.
+
+#include "tip_common.h"
+
+#include <hip/hip_bf16.h>
+
+using bf16 = __hip_bfloat16;
+using short8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+TIP_DEV int swz(int u) { return u ^ ((u >> 4) & 7); }
+
+// Load a 16-B unit (8 bf16) from the swizzled LDS image.
+TIP_DEV short8 lds_read_unit(const short* lds, int u) {
+  return *reinterpret_cast<const short8*>(lds + swz(u) * 8);
+}
+
+TIP_DEV void lds_write_unit(short* lds, int u, short8 v) {
+  *reinterpret_cast<short8*>(lds + swz(u) * 8) = v;
+}
+
+// Stage one NHWC plane [H x W x C] from global into the LDS image
+// [(H+2) x (W+2) x C] interior; halo is zeroed first. 256 threads.
+template <int H, int W, int C>
+TIP_DEV void stage_plane(short* lds, const short* __restrict__ gsrc) {
+  constexpr int UNITS_IMG = (H + 2) * (W + 2) * C / 8;
+  const short8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (int u = threadIdx.x; u < UNITS_IMG; u += blockDim.x) lds_write_unit(lds, u, zero);
+  __syncthreads();
+  constexpr int UNITS = H * W * C / 8;
+  constexpr int UPP = C / 8;  // units per pixel
+  for (int u = threadIdx.x; u < UNITS; u += blockDim.x) {
+    const int p = u / UPP;
+    const int h = u - p * UPP;
+    const int y = p / W, x = p - y * W;
+    const int lu = (((y + 1) * (W + 2)) + (x + 1)) * UPP + h;
+    short8 v = *reinterpret_cast<const short8*>(gsrc + (int64_t)u * 8);
+    lds_write_unit(lds, lu, v);
+  }
+}
+
+// 3x3 conv over the LDS image: computes the [16-pixel x 16-cout] tiles
+// this wave owns and applies bias (+ optional residual from rlds) + relu,
+// then writes either to the out LDS image interior or to global NHWC.
+// K layout: k = tap*C + ci, taps row-major dy,dx in [0,3)x[0,3).
+template <int H, int W, int C, int COUT, bool TO_LDS, bool RESID, int STRIDE>
+TIP_DEV void conv3x3(
+    const short* __restrict__ in_lds,   // [(H+2)(W+2)C] swizzled
+    short* __restrict__ out_lds,        // TO_LDS: [(OH+2)(OW+2)COUT]
+    short* __restrict__ gout,           // !TO_LDS: global NHWC [OH*OW*COUT]
+    const short* __restrict__ rlds,     // RESID: residual LDS image (COUT ch)
+    const short* __restrict__ wpack,    // [ksteps][64][8] per cout-tile:
+                                        // [cout_tiles][ksteps][64][8]
+    const float* __restrict__ bias) {   // [COUT]
+  constexpr int OH = H / STRIDE, OW = W / STRIDE;
+  constexpr int K = 9 * C;
+  constexpr int KSTEPS = (K + 31) / 32;
+  constexpr int NPIX = OH * OW;
+  constexpr int PIX_TILES = NPIX / 16;
+  constexpr int COUT_TILES = COUT / 16;
+  constexpr int UPP = C / 8;
+  constexpr int OUPP = COUT / 8;
+
+  const int lane = lane_id();
+  const int wid = wave_id();
+  const int j = lane & 15;        // output channel within tile
+  const int g = lane >> 4;        // k-group (8 consecutive k)
+
+  // per-(tap-slice) lane constants: which (tap, half-unit) this lane's
+  // 8-element A slice belongs to for each kstep
+  for (int tile = wid; tile < PIX_TILES * COUT_TILES; tile += 4) {
+    const int pt = tile % PIX_TILES;
+    const int ct = tile / PIX_TILES;
+    const int p0 = pt * 16;
+    // this lane's output pixel (for C/D rows) — rows are (g*4 + reg)
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const short* wp = wpack + ((int64_t)ct * KSTEPS + 0) * 64 * 8 + lane * 8;
+#pragma unroll 4
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      // A: lane holds pixels row i = j (col index is the PIXEL for A? no:)
+      // For D[i][j] = sum_k A[i][k] B[k][j]: i = pixel, j = cout.
+      // lane l: A row i = l&15 -> pixel p0 + (l&15); k = g*8+e.
+      const int k0 = ks * 32 + g * 8;
+      short8 a;
+      if (k0 < K) {
+        const int tap = k0 / C;
+        const int ci = k0 - tap * C;  // multiple of 8
+        const int dy = tap / 3, dx = tap - dy * 3;
+        const int pix = p0 + j;       // A row for this lane
+        const int oy = pix / OW, ox = pix - oy * OW;
+        const int iy = oy * STRIDE + dy, ix = ox * STRIDE + dx;  // halo coords
+        const int u = (iy * (W + 2) + ix) * UPP + (ci >> 3);
+        a = lds_read_unit(in_lds, u);
+      } else {
+        a = short8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+      const short8 b = *reinterpret_cast<const short8*>(wp + (int64_t)ks * 64 * 8);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    // epilogue: C/D row = g*4 + reg -> pixel p0 + row; col j = cout ct*16+j
+    const int cout = ct * 16 + j;
+    const float bs = bias[cout];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int pix = p0 + g * 4 + reg;
+      float v = acc[reg] + bs;
+      if (RESID) {
+        const int oy = pix / OW, ox = pix - oy * OW;
+        const int ru = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
+        const short* runit = rlds + swz(ru) * 8;
+        const bf16 rv = reinterpret_cast<const bf16*>(runit)[cout & 7];
+        v += __bfloat162float(rv);
+      }
+      v = fmaxf(v, 0.f);
+      const bf16 ov = __float2bfloat16(v);
+      if (TO_LDS) {
+        const int oy = pix / OW, ox = pix - oy * OW;
+        const int ou = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
+        short* unit = out_lds + swz(ou) * 8;
+        reinterpret_cast<bf16*>(unit)[cout & 7] = ov;
+      } else {
+        reinterpret_cast<bf16*>(gout)[(int64_t)pix * COUT + cout] = ov;
+      }
+    }
+  }
+}
+
+// 1x1 stride-2 shortcut projection, accumulated in the conv2 epilogue is
+// complex; instead we compute the projected residual into rlds first.
+// rlds: [(OH+2) x (OW+2) x COUT] swizzled image (interior only written;
+// halo must be pre-zeroed). wpack1x1: K = C padded to 32.
+template <int H, int W, int C, int COUT>
+TIP_DEV void shortcut1x1_s2(
+    const short* __restrict__ in_lds, short* __restrict__ rlds,
+    const short* __restrict__ wpack, const float* __restrict__ bias) {
+  constexpr int OH = H / 2, OW = W / 2;
+  constexpr int KSTEPS = (C + 31) / 32;
+  constexpr int NPIX = OH * OW;
+  constexpr int PIX_TILES = NPIX / 16;
+  constexpr int COUT_TILES = COUT / 16;
+  constexpr int UPP = C / 8;
+  constexpr int OUPP = COUT / 8;
+  const int lane = lane_id();
+  const int wid = wave_id();
+  const int j = lane & 15;
+  const int g = lane >> 4;
+  for (int tile = wid; tile < PIX_TILES * COUT_TILES; tile += 4) {
+    const int pt = tile % PIX_TILES;
+    const int ct = tile / PIX_TILES;
+    const int p0 = pt * 16;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const short* wp = wpack + ((int64_t)ct * KSTEPS) * 64 * 8 + lane * 8;
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      const int k0 = ks * 32 + g * 8;
+      short8 a = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (k0 < C) {
+        const int pix = p0 + j;
+        const int oy = pix / OW, ox = pix - oy * OW;
+        const int iy = oy * 2 + 1, ix = ox * 2 + 1;  // center tap, halo coords
+        const int u = (iy * (W + 2) + ix) * UPP + (k0 >> 3);
+        a = lds_read_unit(in_lds, u);
+      }
+      const short8 b = *reinterpret_cast<const short8*>(wp + (int64_t)ks * 64 * 8);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    const int cout = ct * 16 + j;
+    const float bs = bias[cout];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int pix = p0 + g * 4 + reg;
+      const int oy = pix / OW, ox = pix - oy * OW;
+      const int ou = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
+      short* unit = rlds + swz(ou) * 8;
+      reinterpret_cast<bf16*>(unit)[cout & 7] = __float2bfloat16(acc[reg] + bs);
+    }
+  }
+}
+
+// ---- kernels ----
+
+// Equal-channel residual block: x -> conv1+relu (LDS) -> conv2+bias+x+relu
+// -> global. One image per workgroup.
+template <int H, int W, int C>
+__launch_bounds__(256) __global__ void resblock_kernel(
+    const short* __restrict__ gin,   // [B, H*W*C] NHWC bf16
+    short* __restrict__ gout,        // [B, H*W*C]
+    const short* __restrict__ w1, const float* __restrict__ b1,
+    const short* __restrict__ w2, const float* __restrict__ b2) {
+  extern __shared__ short lds[];
+  short* bufX = lds;                                // [(H+2)(W+2)C]
+  short* bufH = lds + (H + 2) * (W + 2) * C;        // [(H+2)(W+2)C]
+  const int64_t img_off = (int64_t)blockIdx.x * H * W * C;
+  stage_plane<H, W, C>(bufX, gin + img_off);
+  // zero bufH halo (stage_plane zeroes everything first; emulate)
+  {
+    constexpr int UNITS_IMG = (H + 2) * (W + 2) * C / 8;
+    const short8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int u = threadIdx.x; u < UNITS_IMG; u += blockDim.x)
+      lds_write_unit(bufH, u, zero);
+  }
+  __syncthreads();
+  conv3x3<H, W, C, C, true, false, 1>(bufX, bufH, nullptr, nullptr, w1, b1);
+  __syncthreads();
+  conv3x3<H, W, C, C, false, true, 1>(
+      bufH, nullptr, gout + img_off, bufX, w2, b2);
+}
+
+// Downsample block: conv1 (stride 2, C -> 2C) + relu -> conv2 (2C) +
+// shortcut(1x1 s2) + relu -> global.
+template <int H, int W, int C>
+__launch_bounds__(256) __global__ void downblock_kernel(
+    const short* __restrict__ gin,   // [B, H*W*C]
+    short* __restrict__ gout,        // [B, (H/2)*(W/2)*2C]
+    const short* __restrict__ w1, const float* __restrict__ b1,
+    const short* __restrict__ w2, const float* __restrict__ b2,
+    const short* __restrict__ wsc, const float* __restrict__ bsc) {
+  constexpr int OH = H / 2, OW = W / 2, C2 = 2 * C;
+  extern __shared__ short lds[];
+  short* bufX = lds;                                  // [(H+2)(W+2)C]
+  short* bufH = bufX + (H + 2) * (W + 2) * C;         // [(OH+2)(OW+2)C2]
+  short* bufR = bufH + (OH + 2) * (OW + 2) * C2;      // [(OH+2)(OW+2)C2]
+  const int64_t in_off = (int64_t)blockIdx.x * H * W * C;
+  const int64_t out_off = (int64_t)blockIdx.x * OH * OW * C2;
+  stage_plane<H, W, C>(bufX, gin + in_off);
+  {
+    constexpr int UNITS = ((OH + 2) * (OW + 2) * C2 / 8);
+    const short8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int u = threadIdx.x; u < UNITS; u += blockDim.x) {
+      lds_write_unit(bufH, u, zero);
+      lds_write_unit(bufR, u, zero);
+    }
+  }
+  __syncthreads();
+  conv3x3<H, W, C, C2, true, false, 2>(bufX, bufH, nullptr, nullptr, w1, b1);
+  shortcut1x1_s2<H, W, C, C2>(bufX, bufR, wsc, bsc);
+  __syncthreads();
+  conv3x3<OH, OW, C2, C2, false, true, 1>(
+      bufH, nullptr, gout + out_off, bufR, w2, b2);
+}
+
+// Stem: conv3x3 Cin=8 (4 real + 4 zero-pad channels; 8 keeps units whole)
+// -> 16, relu, to global.
+template <int H, int W, int CIN, int COUT>
+__launch_bounds__(256) __global__ void stem_kernel(
+    const short* __restrict__ gin,   // [B, H*W*CIN]
+    short* __restrict__ gout,        // [B, H*W*COUT]
+    const short* __restrict__ w, const float* __restrict__ b) {
+  extern __shared__ short lds[];
+  short* bufX = lds;
+  const int64_t in_off = (int64_t)blockIdx.x * H * W * CIN;
+  const int64_t out_off = (int64_t)blockIdx.x * H * W * COUT;
+  stage_plane<H, W, CIN>(bufX, gin + in_off);
+  __syncthreads();
+  conv3x3<H, W, CIN, COUT, false, false, 1>(
+      bufX, nullptr, gout + out_off, nullptr, w, b);
+}
+
+// MFMA layout probe: D = A @ B for A [16, 32], B [32, 16] bf16 (row-major)
+// with the exact fragment code paths used above.
+__global__ void mfma_probe_kernel(
+    const short* __restrict__ A, const short* __restrict__ B,
+    float* __restrict__ D) {
+  const int lane = lane_id();
+  const int j = lane & 15, g = lane >> 4;
+  short8 a = *reinterpret_cast<const short8*>(A + (j * 32 + g * 8));
+  short8 b;
+  for (int e = 0; e < 8; ++e) b[e] = B[(g * 8 + e) * 16 + j];
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  for (int reg = 0; reg < 4; ++reg) D[(g * 4 + reg) * 16 + j] = acc[reg];
+}
+
+// ---- launchers ----
+
+void launch_mfma_probe(const short* a, const short* b, float* d, hipStream_t s) {
+  mfma_probe_kernel<<<1, 64, 0, s>>>(a, b, d);
+}
+
+template <int H, int W, int C>
+static void resblock(int batch, const short* gin, short* gout, const short* w1,
+                     const float* b1, const short* w2, const float* b2,
+                     hipStream_t s) {
+  const int lds_bytes = 2 * (H + 2) * (W + 2) * C * 2;
+  auto k = resblock_kernel<H, W, C>;
+  hipFuncSetAttribute((const void*)k,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+  k<<<batch, 256, lds_bytes, s>>>(gin, gout, w1, b1, w2, b2);
+}
+
+void launch_resblock(int variant, int batch, const short* gin, short* gout,
+                     const short* w1, const float* b1, const short* w2,
+                     const float* b2, hipStream_t s) {
+  if (variant == 0) resblock<32, 32, 16>(batch, gin, gout, w1, b1, w2, b2, s);
+  else if (variant == 1) resblock<16, 16, 32>(batch, gin, gout, w1, b1, w2, b2, s);
+  else resblock<8, 8, 64>(batch, gin, gout, w1, b1, w2, b2, s);
+}
+
+template <int H, int W, int C>
+static void downblock(int batch, const short* gin, short* gout,
+                      const short* w1, const float* b1, const short* w2,
+                      const float* b2, const short* wsc, const float* bsc,
+                      hipStream_t s) {
+  constexpr int OH = H / 2, OW = W / 2, C2 = 2 * C;
+  const int lds_bytes =
+      ((H + 2) * (W + 2) * C + 2 * (OH + 2) * (OW + 2) * C2) * 2;
+  auto k = downblock_kernel<H, W, C>;
+  hipFuncSetAttribute((const void*)k,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+  k<<<batch, 256, lds_bytes, s>>>(gin, gout, w1, b1, w2, b2, wsc, bsc);
+}
+
+void launch_downblock(int variant, int batch, const short* gin, short* gout,
+                      const short* w1, const float* b1, const short* w2,
+                      const float* b2, const short* wsc, const float* bsc,
+                      hipStream_t s) {
+  if (variant == 0)
+    downblock<32, 32, 16>(batch, gin, gout, w1, b1, w2, b2, wsc, bsc, s);
+  else
+    downblock<16, 16, 32>(batch, gin, gout, w1, b1, w2, b2, wsc, bsc, s);
+}
+
+void launch_stem(int batch, const short* gin, short* gout, const short* w,
+                 const float* b, hipStream_t s) {
+  constexpr int H = 32, W = 32, CIN = 8, COUT = 16;
+  const int lds_bytes = (H + 2) * (W + 2) * CIN * 2;
+  auto k = stem_kernel<H, W, CIN, COUT>;
+  hipFuncSetAttribute((const void*)k,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+  k<<<batch, 256, lds_bytes, s>>>(gin, gout, w, b);
+}
