@@ -75,6 +75,15 @@ class GraphedExtractor:
         self.batch = batch
         m = fold_bn_inference(model).to(device)
         self.in_ch = 3
+        self.fused = None
+        if device.type == "cuda" and os.environ.get("TIP_NO_FUSED_RESNET") != "1":
+            try:
+                from simple_tip_amd.models.resnet_fused import FusedResNet20
+
+                self.fused = FusedResNet20(m, device)
+            except Exception as e:  # pragma: no cover
+                print(f"[bench] fused ResNet path unavailable ({e!r})",
+                      file=sys.stderr)
         if device.type == "cuda":
             pad_stem_channels(m, 4)  # NHWC bf16 igemm path needs >=4 channels
             self.in_ch = 4
@@ -82,7 +91,7 @@ class GraphedExtractor:
         else:
             self.model = m
         self.graph = None
-        if use_graph and device.type == "cuda":
+        if use_graph and device.type == "cuda" and self.fused is None:
             try:
                 self._capture()
             except Exception as e:  # pragma: no cover - graph capture optional
@@ -124,6 +133,9 @@ class GraphedExtractor:
 
     @torch.no_grad()
     def __call__(self, x):
+        if self.fused is not None:
+            ats, logits = self.fused(x)
+            return ats, torch.softmax(logits, dim=1)
         if self.graph is not None and x.shape[0] == self.batch:
             self.static_x.copy_(self._prep(x))
             self.graph.replay()
