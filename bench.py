@@ -232,6 +232,16 @@ def main():
     )
     train_classes = set(torch.unique(train_pred.cpu()).tolist())
 
+    fused_prio = None
+    if on_gpu and os.environ.get("TIP_NO_FUSED_PRIO") != "1":
+        try:
+            from simple_tip_amd.engine.serving import FusedPrioritizer
+
+            fused_prio = FusedPrioritizer(dsa, lsa, device)
+            log(rank, f"fused prioritizer ready (lsa={fused_prio.lsa_ready})")
+        except Exception as e:  # pragma: no cover
+            log(rank, f"fused prioritizer unavailable: {e!r}")
+
     # pre-generate per-rank test batches (distinct per step and rank, same
     # class-structured distribution as training so predictions spread)
     n_pool = min(max(args.steps + args.warmup, 4), 8)
@@ -266,10 +276,14 @@ def main():
         mark("forward")
         unc = ops.softmax_uncertainties(probs)
         mark("unc")
-        dsa_scores = dsa(ats, pred)
-        mark("dsa")
-        lsa_scores = lsa(ats, pred)
-        mark("lsa")
+        if fused_prio is not None and fused_prio.lsa_ready:
+            dsa_scores, lsa_scores = fused_prio(ats, pred)
+            mark("dsa+lsa")
+        else:
+            dsa_scores = dsa(ats, pred)
+            mark("dsa")
+            lsa_scores = lsa(ats, pred)
+            mark("lsa")
         gini = unc["deep_gini"]
         # publish score shards (tiny, latency-bound on xGMI)
         if world > 1:

@@ -21,6 +21,13 @@ void launch_pairwise_rowmin(const float*, const float*, const float*,
 void launch_pairwise_kde(const float*, const float*, const float*,
                          const float*, int, int, int, float2*, float*,
                          hipStream_t);
+void launch_grouped_rowmin(const float*, const float*, const float*,
+                           const float*, const int*, const int*, int, int,
+                           int, int, float*, int*, float*, int64_t*,
+                           hipStream_t);
+void launch_grouped_kde(const float*, const float*, const float*,
+                        const float*, const int*, const int*, int, int, int,
+                        int, float2*, float*, hipStream_t);
 void launch_profile(int, const float*, const unsigned char*, const float*,
                     const float*, float, int, int, int, int,
                     unsigned long long*, long long*, hipStream_t);
@@ -101,6 +108,47 @@ std::vector<torch::Tensor> rowmin_l2(torch::Tensor a, torch::Tensor b,
                          dist.data_ptr<float>(), idx.data_ptr<int64_t>(),
                          cur_stream());
   return {dist, idx};
+}
+
+std::vector<torch::Tensor> grouped_rowmin(
+    torch::Tensor testS, torch::Tensor trainS, torch::Tensor tseg,
+    torch::Tensor nseg, torch::Tensor bnorm, int64_t jb_max) {
+  check_f32_2d(testS, "testS");
+  check_f32_2d(trainS, "trainS");
+  TORCH_CHECK(tseg.is_cuda() && tseg.dtype() == torch::kInt32);
+  TORCH_CHECK(nseg.is_cuda() && nseg.dtype() == torch::kInt32);
+  const int bp = testS.size(0), k = testS.size(1);
+  const int nclasses = tseg.size(0) - 1;
+  auto an = rownorm(testS);
+  auto pval = torch::empty({jb_max, bp}, testS.options());
+  auto pidx = torch::empty({jb_max, bp}, testS.options().dtype(torch::kInt32));
+  auto dist = torch::empty({bp}, testS.options());
+  auto idx = torch::empty({bp}, testS.options().dtype(torch::kInt64));
+  launch_grouped_rowmin(
+      testS.data_ptr<float>(), trainS.data_ptr<float>(), an.data_ptr<float>(),
+      bnorm.data_ptr<float>(), tseg.data_ptr<int>(), nseg.data_ptr<int>(),
+      nclasses, bp, k, jb_max, pval.data_ptr<float>(), pidx.data_ptr<int>(),
+      dist.data_ptr<float>(), idx.data_ptr<int64_t>(), cur_stream());
+  return {dist, idx};
+}
+
+torch::Tensor grouped_kde(torch::Tensor testWS, torch::Tensor trainWS,
+                          torch::Tensor tseg, torch::Tensor nseg,
+                          torch::Tensor bnorm, int64_t jb_max) {
+  check_f32_2d(testWS, "testWS");
+  check_f32_2d(trainWS, "trainWS");
+  const int bp = testWS.size(0), k = testWS.size(1);
+  const int nclasses = tseg.size(0) - 1;
+  auto an = rownorm(testWS);
+  auto pkde = torch::empty({jb_max, bp, 2}, testWS.options());
+  auto out = torch::empty({bp}, testWS.options());
+  launch_grouped_kde(
+      testWS.data_ptr<float>(), trainWS.data_ptr<float>(),
+      an.data_ptr<float>(), bnorm.data_ptr<float>(), tseg.data_ptr<int>(),
+      nseg.data_ptr<int>(), nclasses, bp, k, jb_max,
+      reinterpret_cast<float2*>(pkde.data_ptr<float>()),
+      out.data_ptr<float>(), cur_stream());
+  return out;
 }
 
 torch::Tensor kde_logsumexp(torch::Tensor test, torch::Tensor train) {
@@ -342,6 +390,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rowmin_l2", &rowmin_l2, py::arg("a"), py::arg("b"),
         py::arg("bnorm") = py::none());
   m.def("kde_logsumexp", &kde_logsumexp);
+  m.def("grouped_rowmin", &grouped_rowmin);
+  m.def("grouped_kde", &grouped_kde);
   m.def("profile", &profile);
   m.def("pack_bits", &pack_bits);
   m.def("popcount_rows", &popcount_rows);

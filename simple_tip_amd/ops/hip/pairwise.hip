@@ -237,6 +237,195 @@ __launch_bounds__(256, 2) __global__ void pairwise_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Grouped (segmented) variant: test rows are sorted by class and padded to
+// 128-row segments (tseg offsets, C+1 entries); train rows are the
+// class-concatenated set with raw offsets (nseg). One launch covers every
+// class — the per-class python loop (10+ launches, ~200 host ops per step)
+// becomes one kernel + one combine. Each block derives its class from its
+// row-block (segments are 128-aligned), and only columns inside the class's
+// train range participate. Argmin indices are GLOBAL train rows, so the
+// precomputed DSA b-table gathers directly.
+// ---------------------------------------------------------------------------
+
+template <int EPI>
+__launch_bounds__(256, 2) __global__ void grouped_pairwise_kernel(
+    const float* __restrict__ A,      // [Bp, K] class-sorted, 128-padded
+    const float* __restrict__ B,      // [Ntot, K] class-concatenated
+    const float* __restrict__ anorm,  // [Bp]
+    const float* __restrict__ bnorm,  // [Ntot]
+    const int* __restrict__ tseg,     // [C+1] padded test offsets (x128)
+    const int* __restrict__ nseg,     // [C+1] train offsets
+    int nclasses, int Bp, int K,
+    float* __restrict__ pmin_val,     // [jb_max, Bp]
+    int* __restrict__ pmin_idx,
+    float2* __restrict__ pkde) {
+  __shared__ float lds[2 * BK * (BM + LDS_PAD)];
+  float* As = lds;
+  float* Bs = lds + BK * (BM + LDS_PAD);
+  __shared__ float red_v[BM][2];
+  __shared__ int red_i[BM][2];
+  __shared__ float red_s[BM][2];
+
+  const int bi = blockIdx.y;
+  const int row0 = bi * BM;
+  // class of this 128-aligned row block
+  int cls = 0;
+  for (int c = 0; c < nclasses; ++c)
+    if (tseg[c] <= row0) cls = c;
+  const int ncol0 = nseg[cls], ncol1 = nseg[cls + 1];
+  const int col0 = ncol0 + blockIdx.x * BN;
+  if (col0 >= ncol1) return;  // this class has fewer column blocks
+  const int bj = blockIdx.x;
+
+  const int lane = lane_id();
+  const int wid = wave_id();
+  const int wr = wid >> 1;
+  const int wc = wid & 1;
+
+  f32x16 acc[2][2] = {};
+  StageRegs ra = stage_load(A, Bp, K, row0, 0);
+  StageRegs rb = stage_load(B, ncol1, K, col0, 0);
+  stage_write(As, ra);
+  stage_write(Bs, rb);
+  __syncthreads();
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    const bool has_next = (k0 + BK) < K;
+    if (has_next) {
+      ra = stage_load(A, Bp, K, row0, k0 + BK);
+      rb = stage_load(B, ncol1, K, col0, k0 + BK);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const int k = kk + (lane >> 5);
+      const float* as = &As[k * (BM + LDS_PAD) + wr * 64 + (lane & 31)];
+      const float* bs = &Bs[k * (BM + LDS_PAD) + wc * 64 + (lane & 31)];
+      const float a0 = as[0], a1 = as[32];
+      const float b0 = bs[0], b1 = bs[32];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+    if (has_next) {
+      stage_write(As, ra);
+      stage_write(Bs, rb);
+    }
+    __syncthreads();
+  }
+
+  const int jl0 = col0 + wc * 64 + (lane & 31);
+  const float bn0 = (jl0 < ncol1) ? bnorm[jl0] : 0.f;
+  const float bn1 = (jl0 + 32 < ncol1) ? bnorm[jl0 + 32] : 0.f;
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int row_local = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+      const int block_row = wr * 64 + m * 32 + row_local;
+      const int i = row0 + block_row;
+      const float an = (i < Bp) ? anorm[i] : 0.f;
+      float d0 = fmaxf(an + bn0 - 2.f * acc[m][0][reg], 0.f);
+      float d1 = fmaxf(an + bn1 - 2.f * acc[m][1][reg], 0.f);
+      if (jl0 >= ncol1) d0 = FLT_MAX;
+      if (jl0 + 32 >= ncol1) d1 = FLT_MAX;
+      if (EPI == EPI_ROWMIN) {
+        MinIdx mi{d0, jl0};
+        mi = min_idx_combine(mi, MinIdx{d1, jl0 + 32});
+        mi = half_reduce_min(mi);
+        if ((lane & 31) == 0) {
+          red_v[block_row][wc] = mi.v;
+          red_i[block_row][wc] = mi.i;
+        }
+      } else {
+        MinIdx mi{d0, 0};
+        mi = min_idx_combine(mi, MinIdx{d1, 0});
+        mi = half_reduce_min(mi);
+        const float tmax = -0.5f * mi.v;
+        float s = 0.f;
+        if (d0 != FLT_MAX) s += __expf(-0.5f * d0 - tmax);
+        if (d1 != FLT_MAX) s += __expf(-0.5f * d1 - tmax);
+        s = half_reduce_sum(s);
+        if ((lane & 31) == 0) {
+          red_v[block_row][wc] = tmax;
+          red_s[block_row][wc] = s;
+        }
+      }
+    }
+  }
+  __syncthreads();
+  for (int r = threadIdx.x; r < BM; r += blockDim.x) {
+    const int i = row0 + r;
+    if (i >= Bp) continue;
+    if (EPI == EPI_ROWMIN) {
+      MinIdx best = min_idx_combine(
+          MinIdx{red_v[r][0], red_i[r][0]}, MinIdx{red_v[r][1], red_i[r][1]});
+      pmin_val[(int64_t)bj * Bp + i] = best.v;
+      pmin_idx[(int64_t)bj * Bp + i] = best.i;
+    } else {
+      float m0 = red_v[r][0], s0 = red_s[r][0];
+      float m1 = red_v[r][1], s1 = red_s[r][1];
+      float mm, ss;
+      if (m0 >= m1) {
+        mm = m0;
+        ss = s0 + ((s1 > 0.f) ? s1 * __expf(m1 - m0) : 0.f);
+      } else {
+        mm = m1;
+        ss = s1 + ((s0 > 0.f) ? s0 * __expf(m0 - m1) : 0.f);
+      }
+      pkde[(int64_t)bj * Bp + i] = make_float2(mm, ss);
+    }
+  }
+}
+
+__global__ void grouped_rowmin_combine_kernel(
+    const float* __restrict__ pval, const int* __restrict__ pidx,
+    const int* __restrict__ tseg, const int* __restrict__ nseg, int nclasses,
+    int Bp, float* __restrict__ out_dist, int64_t* __restrict__ out_idx) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= Bp) return;
+  int cls = 0;
+  for (int c = 0; c < nclasses; ++c)
+    if (tseg[c] <= i) cls = c;
+  const int jcount = (nseg[cls + 1] - nseg[cls] + BN - 1) / BN;
+  MinIdx best{FLT_MAX, 0x7fffffff};
+  for (int b = 0; b < jcount; ++b)
+    best = min_idx_combine(
+        best, MinIdx{pval[(int64_t)b * Bp + i], pidx[(int64_t)b * Bp + i]});
+  if (jcount == 0) {
+    out_dist[i] = FLT_MAX;
+    out_idx[i] = -1;
+  } else {
+    out_dist[i] = sqrtf(best.v);
+    out_idx[i] = best.i;
+  }
+}
+
+__global__ void grouped_kde_combine_kernel(
+    const float2* __restrict__ pkde, const int* __restrict__ tseg,
+    const int* __restrict__ nseg, int nclasses, int Bp,
+    float* __restrict__ out_lse) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= Bp) return;
+  int cls = 0;
+  for (int c = 0; c < nclasses; ++c)
+    if (tseg[c] <= i) cls = c;
+  const int jcount = (nseg[cls + 1] - nseg[cls] + BN - 1) / BN;
+  float mm = -FLT_MAX, ss = 0.f;
+  for (int b = 0; b < jcount; ++b) {
+    const float2 p = pkde[(int64_t)b * Bp + i];
+    if (p.y <= 0.f) continue;
+    if (p.x > mm) {
+      ss = p.y + ((ss > 0.f) ? ss * __expf(mm - p.x) : 0.f);
+      mm = p.x;
+    } else {
+      ss += p.y * __expf(p.x - mm);
+    }
+  }
+  out_lse[i] = (ss > 0.f) ? (mm + __logf(ss)) : -FLT_MAX;
+}
+
 // Deterministic cross-block-column combines (ascending bj keeps np.argmin
 // lowest-index tie semantics; fixed order keeps results bitwise stable).
 __global__ void rowmin_combine_kernel(
@@ -322,4 +511,27 @@ void launch_pairwise_kde(const float* a, const float* b, const float* an,
   pairwise_kernel<EPI_KDE><<<grid, 256, 0, s>>>(
       a, b, an, bn, m, n, k, nullptr, nullptr, nullptr, pkde);
   kde_combine_kernel<<<ceil_div(m, 256), 256, 0, s>>>(pkde, jb, m, out_lse);
+}
+
+void launch_grouped_rowmin(const float* a, const float* b, const float* an,
+                           const float* bn, const int* tseg, const int* nseg,
+                           int nclasses, int bp, int k, int jb_max,
+                           float* pval, int* pidx, float* out_dist,
+                           int64_t* out_idx, hipStream_t s) {
+  dim3 grid(jb_max, ceil_div(bp, BM));
+  grouped_pairwise_kernel<EPI_ROWMIN><<<grid, 256, 0, s>>>(
+      a, b, an, bn, tseg, nseg, nclasses, bp, k, pval, pidx, nullptr);
+  grouped_rowmin_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
+      pval, pidx, tseg, nseg, nclasses, bp, out_dist, out_idx);
+}
+
+void launch_grouped_kde(const float* a, const float* b, const float* an,
+                        const float* bn, const int* tseg, const int* nseg,
+                        int nclasses, int bp, int k, int jb_max, float2* pkde,
+                        float* out_lse, hipStream_t s) {
+  dim3 grid(jb_max, ceil_div(bp, BM));
+  grouped_pairwise_kernel<EPI_KDE><<<grid, 256, 0, s>>>(
+      a, b, an, bn, tseg, nseg, nclasses, bp, k, nullptr, nullptr, pkde);
+  grouped_kde_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
+      pkde, tseg, nseg, nclasses, bp, out_lse);
 }

@@ -1,0 +1,75 @@
+"""FusedPrioritizer (grouped segmented kernels) vs the per-class path."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _fit(n=3000, d=256, classes=7, seed=0):
+    from simple_tip_amd.core.surprise import DSA, LSA, MultiModalSA
+
+    rng = np.random.RandomState(seed)
+    ats = torch.from_numpy(rng.randn(n, d).astype(np.float32))
+    pred = torch.from_numpy(rng.randint(0, classes, n))
+    dev = torch.device("cuda:0")
+    dsa = DSA(ats, pred, device=dev)
+    lsa = MultiModalSA.build_by_class(
+        ats, pred, lambda a, p: LSA(a, max_features=64, device=dev)
+    )
+    return dsa, lsa, dev
+
+
+def test_fused_matches_per_class():
+    from simple_tip_amd.engine.serving import FusedPrioritizer
+
+    dsa, lsa, dev = _fit()
+    fused = FusedPrioritizer(dsa, lsa, dev)
+    assert fused.lsa_ready
+
+    rng = np.random.RandomState(1)
+    test = torch.from_numpy(rng.randn(777, 256).astype(np.float32)).to(dev)
+    tp = torch.from_numpy(rng.randint(0, 7, 777)).to(dev)
+
+    d_fused, l_fused = fused(test, tp)
+    d_ref = dsa(test, tp)
+    l_ref = lsa(test, tp)
+
+    assert torch.allclose(d_fused.cpu(), d_ref.cpu(), rtol=1e-4, atol=1e-5)
+    assert torch.allclose(
+        l_fused.cpu().double(), l_ref.cpu().double(), rtol=1e-3, atol=1e-3
+    )
+
+
+def test_fused_skewed_classes():
+    from simple_tip_amd.engine.serving import FusedPrioritizer
+
+    dsa, lsa, dev = _fit(seed=2)
+    fused = FusedPrioritizer(dsa, lsa, dev)
+    rng = np.random.RandomState(3)
+    test = torch.from_numpy(rng.randn(500, 256).astype(np.float32)).to(dev)
+    # heavy skew incl. an absent class
+    tp = torch.from_numpy(
+        np.concatenate([np.full(450, 3), rng.randint(0, 2, 50)])
+    ).to(dev)
+    d_fused, l_fused = fused(test, tp)
+    d_ref = dsa(test, tp)
+    assert torch.allclose(d_fused.cpu(), d_ref.cpu(), rtol=1e-4, atol=1e-5)
+    l_ref = lsa(test, tp)
+    assert torch.allclose(
+        l_fused.cpu().double(), l_ref.cpu().double(), rtol=1e-3, atol=1e-3
+    )
+
+
+def test_fused_determinism():
+    from simple_tip_amd.engine.serving import FusedPrioritizer
+
+    dsa, lsa, dev = _fit(seed=4)
+    fused = FusedPrioritizer(dsa, lsa, dev)
+    rng = np.random.RandomState(5)
+    test = torch.from_numpy(rng.randn(300, 256).astype(np.float32)).to(dev)
+    tp = torch.from_numpy(rng.randint(0, 7, 300)).to(dev)
+    d1, l1 = fused(test, tp)
+    d2, l2 = fused(test, tp)
+    assert torch.equal(d1, d2) and torch.equal(l1, l2)
