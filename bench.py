@@ -156,7 +156,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--batch", type=int, default=10240, help="test inputs per GPU per step")
     ap.add_argument("--train-n", type=int, default=TRAIN_N)
-    ap.add_argument("--setup-epochs", type=int, default=2,
+    ap.add_argument("--setup-epochs", type=int, default=4,
                     help="untimed warm-up training epochs (class diversity)")
     args = ap.parse_args()
 

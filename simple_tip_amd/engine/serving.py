@@ -59,47 +59,57 @@ class FusedPrioritizer:
             ),
         )
 
-        # per-class LSA state: kept columns, whitening L^-T, whitened train
+        # per-class LSA state. A class is either
+        #   "fused":    (keep idx, L^-T, const) + its whitened-train segment
+        #   "const":    degraded KDE (reference semantics: density-0 classes
+        #               score 0; prepare-failed classes score +inf)
+        #   "fallback": feature count differs from the majority (drop-feature
+        #               retry ladder fired) — scored via the per-class SA
         self.lsa_ready = False
         if lsa is not None:
-            keeps, linvs, wtrains, woffs, consts = [], [], [], [0], []
+            self.lsa_mode = []
+            wtrains, woffs = [], [0]
+            consts = [0.0] * self.num_classes
             d_ref = None
-            ok = True
+            any_fused = False
             for c in range(self.num_classes):
                 sa = lsa.modal_sa.get(c)
-                if (
-                    sa is None
-                    or not isinstance(sa, LSA)
-                    or sa.kde is None
-                    or sa.kde.prepare_failed
-                ):
-                    ok = False
-                    break
+                if sa is None or not isinstance(sa, LSA):
+                    self.lsa_mode.append(("const", float("inf")))
+                    woffs.append(woffs[-1])
+                    continue
+                if sa.kde is None:
+                    # all features removed / singleton class: density 0
+                    self.lsa_mode.append(("const", 0.0))
+                    woffs.append(woffs[-1])
+                    continue
+                if sa.kde.prepare_failed:
+                    self.lsa_mode.append(("const", float("inf")))
+                    woffs.append(woffs[-1])
+                    continue
                 linv_t, xw, const = sa.kde.device_state(device)
-                keeps.append(
+                if d_ref is None:
+                    d_ref = linv_t.shape[0]
+                if linv_t.shape[0] != d_ref:
+                    self.lsa_mode.append(("fallback", sa))
+                    woffs.append(woffs[-1])
+                    continue
+                keep = (
                     torch.from_numpy(
                         np.delete(
-                            np.arange(
-                                linv_t.shape[0] + len(sa.removed_neurons)
-                            ),
+                            np.arange(linv_t.shape[0] + len(sa.removed_neurons)),
                             sa.removed_neurons,
                         )
                     ).to(device)
                     if sa.removed_neurons
                     else None
                 )
-                linvs.append(linv_t)
+                self.lsa_mode.append(("fused", keep, linv_t))
                 wtrains.append(xw)
                 woffs.append(woffs[-1] + xw.shape[0])
-                consts.append(const)
-                if d_ref is None:
-                    d_ref = linv_t.shape[0]
-                elif linv_t.shape[0] != d_ref:
-                    ok = False
-                    break
-            if ok and len(linvs) == self.num_classes:
-                self.lsa_keep = keeps
-                self.lsa_linv = linvs  # per class [d, d] (L^-T)
+                consts[c] = const
+                any_fused = True
+            if any_fused:
                 self.lsa_wtrainS = torch.cat(wtrains).contiguous()
                 self.lsa_wnormS = (
                     self.lsa_wtrainS * self.lsa_wtrainS
@@ -108,10 +118,16 @@ class FusedPrioritizer:
                     woffs, dtype=torch.int32, device=device
                 )
                 self.lsa_jb_max = max(
-                    (woffs[c + 1] - woffs[c] + 127) // 128
-                    for c in range(self.num_classes)
+                    1,
+                    max(
+                        (woffs[c + 1] - woffs[c] + 127) // 128
+                        for c in range(self.num_classes)
+                    ),
                 )
-                self.lsa_consts = torch.tensor(consts, device=device)
+                self.lsa_d = d_ref
+                self.lsa_consts = torch.tensor(
+                    consts, dtype=torch.float32, device=device
+                )
                 self.lsa_ready = True
 
     def _segment(self, ats: torch.Tensor, pred: torch.Tensor):
@@ -159,16 +175,16 @@ class FusedPrioritizer:
 
         lsa = None
         if self.lsa_ready:
-            d = self.lsa_linv[0].shape[0]
-            white = torch.zeros(bp, d, device=ats.device)
+            white = torch.zeros(bp, self.lsa_d, device=ats.device)
             for c in range(self.num_classes):
                 lo, hi = int(tseg_cpu[c]), int(tseg_cpu[c + 1])
-                if hi <= lo:
+                if hi <= lo or self.lsa_mode[c][0] != "fused":
                     continue
+                _, keep, linv_t = self.lsa_mode[c]
                 seg = padded[lo:hi]
-                if self.lsa_keep[c] is not None:
-                    seg = seg.index_select(1, self.lsa_keep[c])
-                white[lo:hi] = seg @ self.lsa_linv[c]
+                if keep is not None:
+                    seg = seg.index_select(1, keep)
+                white[lo:hi] = seg @ linv_t
             lse = self.ext.grouped_kde(
                 white.contiguous(), self.lsa_wtrainS, tseg, self.lsa_nseg,
                 self.lsa_wnormS, self.lsa_jb_max,
@@ -176,7 +192,21 @@ class FusedPrioritizer:
             cls_of_row = torch.bucketize(
                 torch.arange(bp, device=ats.device), tseg.long()[1:], right=True
             )
-            lsa_sorted = -(lse + self.lsa_consts.float()[cls_of_row])
+            lsa_sorted = -(lse + self.lsa_consts[cls_of_row])
+            # patch degraded / fallback classes
+            for c in range(self.num_classes):
+                mode = self.lsa_mode[c]
+                if mode[0] == "fused":
+                    continue
+                lo, hi = int(tseg_cpu[c]), int(tseg_cpu[c + 1])
+                if hi <= lo:
+                    continue
+                if mode[0] == "const":
+                    lsa_sorted[lo:hi] = mode[1]
+                else:  # fallback: per-class SA on the real rows
+                    lsa_sorted[lo:hi] = mode[1](padded[lo:hi]).float().to(
+                        ats.device
+                    )
             lsa = torch.empty(b, device=ats.device)
             lsa[order] = lsa_sorted[dest]
         return dsa, lsa

@@ -62,6 +62,32 @@ def test_fused_skewed_classes():
     )
 
 
+def test_fused_degraded_classes():
+    """Classes whose KDE can't fit (singleton class) degrade to the
+    reference's constant scores, same as the per-class path."""
+    from simple_tip_amd.core.surprise import DSA, LSA, MultiModalSA
+    from simple_tip_amd.engine.serving import FusedPrioritizer
+
+    rng = np.random.RandomState(7)
+    ats = torch.from_numpy(rng.randn(400, 64).astype(np.float32))
+    pred = torch.from_numpy(rng.randint(0, 3, 400))
+    pred[0] = 3  # class 3 has exactly one training sample
+    dev = torch.device("cuda:0")
+    dsa = DSA(ats, pred, device=dev)
+    lsa = MultiModalSA.build_by_class(
+        ats, pred, lambda a, p: LSA(a, max_features=32, device=dev)
+    )
+    fused = FusedPrioritizer(dsa, lsa, dev)
+    assert fused.lsa_ready
+    test = torch.from_numpy(rng.randn(100, 64).astype(np.float32)).to(dev)
+    tp = torch.from_numpy(rng.randint(0, 4, 100)).to(dev)
+    d_f, l_f = fused(test, tp)
+    l_ref = lsa(test, tp)
+    assert torch.allclose(
+        l_f.cpu().double(), l_ref.cpu().double(), rtol=1e-3, atol=1e-3
+    )
+
+
 def test_fused_determinism():
     from simple_tip_amd.engine.serving import FusedPrioritizer
 
