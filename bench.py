@@ -91,13 +91,36 @@ class GraphedExtractor:
         else:
             self.model = m
         self.graph = None
-        if use_graph and device.type == "cuda" and self.fused is None:
+        self.fused_graph = None
+        if use_graph and device.type == "cuda":
             try:
-                self._capture()
+                if self.fused is not None:
+                    self._capture_fused()
+                else:
+                    self._capture()
             except Exception as e:  # pragma: no cover - graph capture optional
                 print(f"[bench] hipGraph capture failed ({e!r}); eager path",
                       file=sys.stderr)
                 self.graph = None
+                self.fused_graph = None
+
+    @torch.no_grad()
+    def _capture_fused(self):
+        """Capture the whole fused forward (NHWC conversion + 11 block
+        kernels + pool/fc + softmax) into one hipGraph replay."""
+        self.fstatic_x = torch.zeros(self.batch, 3, 32, 32, device=self.device)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self.fused(self.fstatic_x)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            ats, logits = self.fused(self.fstatic_x)
+            self.fstatic_ats = ats
+            self.fstatic_probs = torch.softmax(logits, dim=1)
+        self.fused_graph = g
 
     @torch.no_grad()
     def _capture(self):
@@ -134,6 +157,10 @@ class GraphedExtractor:
     @torch.no_grad()
     def __call__(self, x):
         if self.fused is not None:
+            if self.fused_graph is not None and x.shape[0] == self.batch:
+                self.fstatic_x.copy_(x.to(self.device, torch.float32))
+                self.fused_graph.replay()
+                return self.fstatic_ats, self.fstatic_probs
             ats, logits = self.fused(x)
             return ats, torch.softmax(logits, dim=1)
         if self.graph is not None and x.shape[0] == self.batch:

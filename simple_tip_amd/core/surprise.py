@@ -504,6 +504,11 @@ class DSA(SA):
         dev = self.train_activations.device
         target_ats = target_ats.to(dev, self.train_activations.dtype)
         target_pred = target_pred.to(dev)
+        if target_ats.is_cuda:
+            # grouped fast path: one segmented kernel launch for all classes
+            fused = self._grouped_call(target_ats, target_pred)
+            if fused is not None:
+                return fused
         dsa = torch.empty(target_pred.shape[0], dtype=target_ats.dtype, device=dev)
         if self._class_cache is None:
             self._build_class_cache()
@@ -530,4 +535,25 @@ class DSA(SA):
                 dsa[sel] = 0.0  # single-class training set: no contrast
                 continue
             dsa[sel] = dist_a / b_table[closest_idx]
+        return dsa
+
+    def _grouped_call(self, target_ats, target_pred):
+        """Single-launch DSA via the grouped segmented kernel (GPU only).
+
+        Returns None when inapplicable (labels outside the fitted range);
+        numerics equal the per-class path (same kernels, same tie rules)."""
+        try:
+            from ..engine.serving import FusedPrioritizer
+        except ImportError:  # pragma: no cover
+            return None
+        if bool((target_pred >= self.num_classes).any()):
+            return None
+        fp = getattr(self, "_fused_prio", None)
+        if fp is None:
+            try:
+                fp = FusedPrioritizer(self, None, target_ats.device)
+            except Exception:  # noqa: BLE001 - fall back to per-class loop
+                return None
+            self._fused_prio = fp
+        dsa, _ = fp(target_ats.float(), target_pred)
         return dsa
