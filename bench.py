@@ -106,18 +106,18 @@ class GraphedExtractor:
 
     @torch.no_grad()
     def _capture_fused(self):
-        """Capture the whole fused forward (NHWC conversion + 11 block
-        kernels + pool/fc + softmax) into one hipGraph replay."""
-        self.fstatic_x = torch.zeros(self.batch, 3, 32, 32, device=self.device)
+        """Capture the whole fused forward (channel pad + 11 block kernels +
+        pool/fc + softmax) into one hipGraph replay. Input is NHWC fp32."""
+        self.fstatic_x = torch.zeros(self.batch, 32, 32, 3, device=self.device)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
-                self.fused(self.fstatic_x)
+                self.fused.forward_nhwc(self.fstatic_x)
         torch.cuda.current_stream().wait_stream(s)
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            ats, logits = self.fused(self.fstatic_x)
+            ats, logits = self.fused.forward_nhwc(self.fstatic_x)
             self.fstatic_ats = ats
             self.fstatic_probs = torch.softmax(logits, dim=1)
         self.fused_graph = g
@@ -157,11 +157,18 @@ class GraphedExtractor:
     @torch.no_grad()
     def __call__(self, x):
         if self.fused is not None:
-            if self.fused_graph is not None and x.shape[0] == self.batch:
+            is_nhwc = x.dim() == 4 and x.shape[-1] == 3
+            if (
+                self.fused_graph is not None
+                and is_nhwc
+                and x.shape[0] == self.batch
+            ):
                 self.fstatic_x.copy_(x.to(self.device, torch.float32))
                 self.fused_graph.replay()
                 return self.fstatic_ats, self.fstatic_probs
-            ats, logits = self.fused(x)
+            ats, logits = (
+                self.fused.forward_nhwc(x) if is_nhwc else self.fused(x)
+            )
             return ats, torch.softmax(logits, dim=1)
         if self.graph is not None and x.shape[0] == self.batch:
             self.static_x.copy_(self._prep(x))
@@ -276,11 +283,19 @@ def main():
     # 125 MB H2D transfer overlaps the compute instead of serialising it.
     n_pool = min(max(args.steps + args.warmup, 4), 8)
     pool, labels_pool = [], []
+    nhwc_pool = extractor.fused is not None
+    from simple_tip_amd.studies.synthetic import corrupt_images
     for i in range(n_pool):
         px, py = synthetic_images(
             "bench_cifar10", f"test-r{rank}-{i}", args.batch, (3, 32, 32), 10
         )
+        # reference OOD recipe: half the batch is corrupted, so the model
+        # has real faults for the APFD quality signal
+        half = args.batch // 2
+        px[half:] = corrupt_images("bench_cifar10", px[half:], severity=0.5)
         t = torch.from_numpy(px)
+        if nhwc_pool:
+            t = t.permute(0, 2, 3, 1).contiguous()
         if on_gpu:
             t = t.pin_memory()
         pool.append(t)
@@ -288,9 +303,10 @@ def main():
 
     if on_gpu:
         copy_stream = torch.cuda.Stream()
-        dev_bufs = [
-            torch.empty(args.batch, 3, 32, 32, device=device) for _ in range(2)
-        ]
+        buf_shape = (
+            (args.batch, 32, 32, 3) if nhwc_pool else (args.batch, 3, 32, 32)
+        )
+        dev_bufs = [torch.empty(*buf_shape, device=device) for _ in range(2)]
         copy_events = [torch.cuda.Event(), torch.cuda.Event()]
         consumed_events = [torch.cuda.Event(), torch.cuda.Event()]
 

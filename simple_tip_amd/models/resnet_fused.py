@@ -90,11 +90,17 @@ class FusedResNet20:
 
     @torch.no_grad()
     def __call__(self, x_nchw: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-        b = x_nchw.shape[0]
         x = x_nchw.to(self.device, torch.float32)
-        # NCHW -> NHWC, pad channels to 8
+        return self.forward_nhwc(x.permute(0, 2, 3, 1))
+
+    @torch.no_grad()
+    def forward_nhwc(self, x_nhwc: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Forward for NHWC inputs [B, 32, 32, C<=8] (skips the 125-MB
+        strided NCHW->NHWC transpose on the hot path)."""
+        b = x_nhwc.shape[0]
+        x = x_nhwc.to(self.device)
         nhwc = torch.zeros(b, 32, 32, 8, device=self.device, dtype=torch.bfloat16)
-        nhwc[..., : x.shape[1]] = x.permute(0, 2, 3, 1).to(torch.bfloat16)
+        nhwc[..., : x.shape[-1]] = x.to(torch.bfloat16)
         cur = self.ext.resnet_stem(
             nhwc.reshape(b, -1).contiguous(), self.stem_w.reshape(-1, 8),
             self.stem_b,
