@@ -89,31 +89,33 @@ TIP_DEV void conv3x3(
   const int wid = wave_id();
   const int j = lane & 15;        // output channel within tile
   const int g = lane >> 4;        // k-group (8 consecutive k)
+  constexpr int RS = (W + 2) * UPP;  // LDS units per halo row
 
-  // per-(tap-slice) lane constants: which (tap, half-unit) this lane's
-  // 8-element A slice belongs to for each kstep
   for (int tile = wid; tile < PIX_TILES * COUT_TILES; tile += 4) {
     const int pt = tile % PIX_TILES;
     const int ct = tile / PIX_TILES;
     const int p0 = pt * 16;
-    // this lane's output pixel (for C/D rows) — rows are (g*4 + reg)
+    // A-side pixel for this lane (MFMA row i = lane&15); base LDS unit of
+    // its top-left tap, hoisted out of the K loop
+    const int apix = p0 + j;
+    const int aoy = apix / OW, aox = apix - aoy * OW;
+    const int abase = (aoy * STRIDE) * RS + (aox * STRIDE) * UPP;
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     const short* wp = wpack + ((int64_t)ct * KSTEPS + 0) * 64 * 8 + lane * 8;
-#pragma unroll 4
+#pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
-      // A: lane holds pixels row i = j (col index is the PIXEL for A? no:)
-      // For D[i][j] = sum_k A[i][k] B[k][j]: i = pixel, j = cout.
-      // lane l: A row i = l&15 -> pixel p0 + (l&15); k = g*8+e.
+      // For D[i][j] = sum_k A[i][k] B[k][j]: i = pixel, j = cout;
+      // lane l supplies A[i = l&15][k = g*8 + e]. k decomposes to
+      // (tap, channel) with constexpr C a power of two, so tap/ci are
+      // shifts; dy = tap/3 via mul-shift (tap < 10).
       const int k0 = ks * 32 + g * 8;
       short8 a;
       if (k0 < K) {
-        const int tap = k0 / C;
-        const int ci = k0 - tap * C;  // multiple of 8
-        const int dy = tap / 3, dx = tap - dy * 3;
-        const int pix = p0 + j;       // A row for this lane
-        const int oy = pix / OW, ox = pix - oy * OW;
-        const int iy = oy * STRIDE + dy, ix = ox * STRIDE + dx;  // halo coords
-        const int u = (iy * (W + 2) + ix) * UPP + (ci >> 3);
+        const int tap = k0 / C;       // shift (C is a constexpr power of 2)
+        const int ci = k0 & (C - 1);
+        const int dy = (tap * 11) >> 5;  // == tap/3 for tap in [0, 9]
+        const int dx = tap - dy * 3;
+        const int u = abase + dy * RS + dx * UPP + (ci >> 3);
         a = lds_read_unit(in_lds, u);
       } else {
         a = short8{0, 0, 0, 0, 0, 0, 0, 0};
