@@ -91,61 +91,70 @@ TIP_DEV void conv3x3(
   const int g = lane >> 4;        // k-group (8 consecutive k)
   constexpr int RS = (W + 2) * UPP;  // LDS units per halo row
 
-  for (int tile = wid; tile < PIX_TILES * COUT_TILES; tile += 4) {
-    const int pt = tile % PIX_TILES;
-    const int ct = tile / PIX_TILES;
-    const int p0 = pt * 16;
-    // A-side pixel for this lane (MFMA row i = lane&15); base LDS unit of
-    // its top-left tap, hoisted out of the K loop
-    const int apix = p0 + j;
-    const int aoy = apix / OW, aox = apix - aoy * OW;
-    const int abase = (aoy * STRIDE) * RS + (aox * STRIDE) * UPP;
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    const short* wp = wpack + ((int64_t)ct * KSTEPS + 0) * 64 * 8 + lane * 8;
+  // cout-tile outer loop: the tile's weight fragments are preloaded into
+  // registers ONCE (KSTEPS x 16 B per lane) so the MFMA loop is pure
+  // ds_read + mfma — no global loads on the critical path (the per-mfma
+  // L2 weight load was the previous bound at ~2 waves/SIMD occupancy).
+  for (int ct = 0; ct < COUT_TILES; ++ct) {
+    short8 bfrag[KSTEPS];
+    {
+      const short* wp = wpack + ((int64_t)ct * KSTEPS) * 64 * 8 + lane * 8;
 #pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) {
-      // For D[i][j] = sum_k A[i][k] B[k][j]: i = pixel, j = cout;
-      // lane l supplies A[i = l&15][k = g*8 + e]. k decomposes to
-      // (tap, channel) with constexpr C a power of two, so tap/ci are
-      // shifts; dy = tap/3 via mul-shift (tap < 10).
-      const int k0 = ks * 32 + g * 8;
-      short8 a;
-      if (k0 < K) {
-        const int tap = k0 / C;       // shift (C is a constexpr power of 2)
-        const int ci = k0 & (C - 1);
-        const int dy = (tap * 11) >> 5;  // == tap/3 for tap in [0, 9]
-        const int dx = tap - dy * 3;
-        const int u = abase + dy * RS + dx * UPP + (ci >> 3);
-        a = lds_read_unit(in_lds, u);
-      } else {
-        a = short8{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-      const short8 b = *reinterpret_cast<const short8*>(wp + (int64_t)ks * 64 * 8);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+      for (int ks = 0; ks < KSTEPS; ++ks)
+        bfrag[ks] = *reinterpret_cast<const short8*>(wp + (int64_t)ks * 64 * 8);
     }
-    // epilogue: C/D row = g*4 + reg -> pixel p0 + row; col j = cout ct*16+j
     const int cout = ct * 16 + j;
     const float bs = bias[cout];
+    for (int pt = wid; pt < PIX_TILES; pt += 4) {
+      const int p0 = pt * 16;
+      // A-side pixel for this lane (MFMA row i = lane&15); base LDS unit
+      // of its top-left tap, hoisted out of the K loop
+      const int apix = p0 + j;
+      const int aoy = apix / OW, aox = apix - aoy * OW;
+      const int abase = (aoy * STRIDE) * RS + (aox * STRIDE) * UPP;
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      const int pix = p0 + g * 4 + reg;
-      float v = acc[reg] + bs;
-      if (RESID) {
-        const int oy = pix / OW, ox = pix - oy * OW;
-        const int ru = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
-        const short* runit = rlds + swz(ru) * 8;
-        const bf16 rv = reinterpret_cast<const bf16*>(runit)[cout & 7];
-        v += __bfloat162float(rv);
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        // For D[i][j] = sum_k A[i][k] B[k][j]: i = pixel, j = cout;
+        // lane l supplies A[i = l&15][k = g*8 + e]. k decomposes to
+        // (tap, channel) with constexpr power-of-two C, so tap/ci are
+        // shifts; dy = tap/3 via mul-shift (tap < 10).
+        const int k0 = ks * 32 + g * 8;
+        short8 a;
+        if (k0 < K) {
+          const int tap = k0 / C;
+          const int ci = k0 & (C - 1);
+          const int dy = (tap * 11) >> 5;  // == tap/3 for tap in [0, 9]
+          const int dx = tap - dy * 3;
+          const int u = abase + dy * RS + dx * UPP + (ci >> 3);
+          a = lds_read_unit(in_lds, u);
+        } else {
+          a = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfrag[ks], acc, 0, 0, 0);
       }
-      v = fmaxf(v, 0.f);
-      const bf16 ov = __float2bfloat16(v);
-      if (TO_LDS) {
-        const int oy = pix / OW, ox = pix - oy * OW;
-        const int ou = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
-        short* unit = out_lds + swz(ou) * 8;
-        reinterpret_cast<bf16*>(unit)[cout & 7] = ov;
-      } else {
-        reinterpret_cast<bf16*>(gout)[(int64_t)pix * COUT + cout] = ov;
+      // epilogue: C/D row = g*4 + reg -> pixel p0 + row; col j = cout
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int pix = p0 + g * 4 + reg;
+        float v = acc[reg] + bs;
+        if (RESID) {
+          const int oy = pix / OW, ox = pix - oy * OW;
+          const int ru = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
+          const short* runit = rlds + swz(ru) * 8;
+          const bf16 rv = reinterpret_cast<const bf16*>(runit)[cout & 7];
+          v += __bfloat162float(rv);
+        }
+        v = fmaxf(v, 0.f);
+        const bf16 ov = __float2bfloat16(v);
+        if (TO_LDS) {
+          const int oy = pix / OW, ox = pix - oy * OW;
+          const int ou = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
+          short* unit = out_lds + swz(ou) * 8;
+          reinterpret_cast<bf16*>(unit)[cout & 7] = ov;
+        } else {
+          reinterpret_cast<bf16*>(gout)[(int64_t)pix * COUT + cout] = ov;
+        }
       }
     }
   }
