@@ -105,38 +105,27 @@ TIP_DEV void conv3x3(
     }
     const int cout = ct * 16 + j;
     const float bs = bias[cout];
-    for (int pt = wid; pt < PIX_TILES; pt += 4) {
-      const int p0 = pt * 16;
-      // A-side pixel for this lane (MFMA row i = lane&15); base LDS unit
-      // of its top-left tap, hoisted out of the K loop
-      const int apix = p0 + j;
+
+    // LDS unit offset of k-step ks relative to a pixel's top-left tap
+    // (constexpr after full unroll): k decomposes into (tap, channel) with
+    // power-of-two C, dy = tap/3 via mul-shift (tap < 10).
+    auto a_off = [&](int ks) {
+      const int k0 = ks * 32 + g * 8;
+      const int tap = k0 / C;
+      const int ci = k0 & (C - 1);
+      const int dy = (tap * 11) >> 5;
+      const int dx = tap - dy * 3;
+      return dy * RS + dx * UPP + (ci >> 3);
+    };
+    auto a_base = [&](int pt) {
+      const int apix = pt * 16 + j;
       const int aoy = apix / OW, aox = apix - aoy * OW;
-      const int abase = (aoy * STRIDE) * RS + (aox * STRIDE) * UPP;
-      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int ks = 0; ks < KSTEPS; ++ks) {
-        // For D[i][j] = sum_k A[i][k] B[k][j]: i = pixel, j = cout;
-        // lane l supplies A[i = l&15][k = g*8 + e]. k decomposes to
-        // (tap, channel) with constexpr power-of-two C, so tap/ci are
-        // shifts; dy = tap/3 via mul-shift (tap < 10).
-        const int k0 = ks * 32 + g * 8;
-        short8 a;
-        if (k0 < K) {
-          const int tap = k0 / C;
-          const int ci = k0 & (C - 1);
-          const int dy = (tap * 11) >> 5;  // == tap/3 for tap in [0, 9]
-          const int dx = tap - dy * 3;
-          const int u = abase + dy * RS + dx * UPP + (ci >> 3);
-          a = lds_read_unit(in_lds, u);
-        } else {
-          a = short8{0, 0, 0, 0, 0, 0, 0, 0};
-        }
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfrag[ks], acc, 0, 0, 0);
-      }
-      // epilogue: C/D row = g*4 + reg -> pixel p0 + row; col j = cout
+      return (aoy * STRIDE) * RS + (aox * STRIDE) * UPP;
+    };
+    auto epilogue = [&](int pt, const f32x4& acc) {
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int pix = p0 + g * 4 + reg;
+        const int pix = pt * 16 + g * 4 + reg;
         float v = acc[reg] + bs;
         if (RESID) {
           const int oy = pix / OW, ox = pix - oy * OW;
@@ -156,6 +145,49 @@ TIP_DEV void conv3x3(
           reinterpret_cast<bf16*>(gout)[(int64_t)pix * COUT + cout] = ov;
         }
       }
+    };
+
+    // Two independent pixel tiles per iteration: two MFMA accumulation
+    // chains interleave, hiding the dependent-accumulator and LDS-read
+    // latency that a single chain serialises on.
+    int pt = wid;
+    for (; pt + 4 < PIX_TILES; pt += 8) {
+      const int b0 = a_base(pt), b1 = a_base(pt + 4);
+      f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+      f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        const int k0 = ks * 32 + g * 8;
+        short8 a0, a1;
+        if (k0 < K) {
+          const int off = a_off(ks);
+          a0 = lds_read_unit(in_lds, b0 + off);
+          a1 = lds_read_unit(in_lds, b1 + off);
+        } else {
+          a0 = short8{0, 0, 0, 0, 0, 0, 0, 0};
+          a1 = a0;
+        }
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag[ks], acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag[ks], acc1, 0, 0, 0);
+      }
+      epilogue(pt, acc0);
+      epilogue(pt + 4, acc1);
+    }
+    for (; pt < PIX_TILES; pt += 4) {
+      const int b0 = a_base(pt);
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        const int k0 = ks * 32 + g * 8;
+        short8 a;
+        if (k0 < K) {
+          a = lds_read_unit(in_lds, b0 + a_off(ks));
+        } else {
+          a = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfrag[ks], acc, 0, 0, 0);
+      }
+      epilogue(pt, acc);
     }
   }
 }
