@@ -78,6 +78,20 @@ def test_surprise_handler_gpu_end_to_end():
     )
 
 
+def test_active_learning_gpu(tmp_path, monkeypatch):
+    import simple_tip_amd.config as config
+    from simple_tip_amd.studies import get_case_study
+
+    monkeypatch.setattr(config, "OUTPUT_FOLDER", str(tmp_path))
+    study = get_case_study("mnist", scale=0.004, device=torch.device("cuda:0"))
+    study.train([1])
+    study.run_active_learning_eval([1])
+    al = {f.name for f in (tmp_path / "active_learning").iterdir()}
+    assert "mnist_1_original_na.pickle" in al
+    assert "mnist_1_random_nominal.pickle" in al
+    assert "mnist_1_dsa-cam_ood.pickle" in al
+
+
 def test_prio_eval_gpu_artifacts(tmp_path, monkeypatch):
     import simple_tip_amd.config as config
     from simple_tip_amd.studies import get_case_study
