@@ -253,9 +253,19 @@ class MultiModalSA(SA):
             try:
                 sa = self.modal_sa[int(modal_id)]
             except KeyError:
-                raise ValueError(
-                    f"No modal found for modal id {modal_id}. Check your discriminator"
+                # Sample routed to a mode that never occurred in the fitting
+                # data (e.g. a predicted class absent from the training
+                # predictions). The reference raises here
+                # (surprise.py:308-315); we treat such inputs as maximally
+                # surprising instead, consistent with DSA's handling.
+                warnings.warn(
+                    f"No modal found for modal id {modal_id}; scoring those "
+                    f"inputs as maximally surprising (+inf).",
+                    UserWarning,
                 )
+                sel = modal_idx == modal_id
+                res[sel.to(res.device)] = float("inf")
+                continue
             sel = modal_idx == modal_id
             a = acts[sel.to(acts.device)]
             p = None if preds is None else preds[sel.to(preds.device)]
