@@ -410,12 +410,20 @@ def main():
     gini, _, _, pred = last
     last_labels = labels_pool[(args.warmup + args.steps - 1) % len(pool)]
     mis = (pred.cpu().numpy() != last_labels.numpy())
-    apfd = (
-        apfd_from_order(mis, np.argsort(-gini.cpu().numpy(), kind="stable"))
-        if mis.any()
-        else float("nan")
-    )
-    log(rank, f"last-batch accuracy={1.0 - mis.mean():.3f} apfd_gini={apfd:.3f}")
+    def _apfd(scores):
+        if not mis.any():
+            return float("nan")
+        return apfd_from_order(
+            mis, np.argsort(-scores.float().cpu().numpy(), kind="stable")
+        )
+
+    apfd = _apfd(gini)
+    _, dsa_last, lsa_last, _ = last
+    apfd_dsa = _apfd(dsa_last)
+    apfd_lsa = _apfd(lsa_last) if lsa_last is not None else float("nan")
+    log(rank, f"last-batch accuracy={1.0 - mis.mean():.3f} "
+              f"apfd_gini={apfd:.3f} apfd_dsa={apfd_dsa:.3f} "
+              f"apfd_lsa={apfd_lsa:.3f}")
 
     if rank == 0:
         print(
@@ -442,6 +450,8 @@ def main():
                         "at_width": 4096,
                         "scorers": "gini+softmax-family+dsa+pc-lsa",
                         "apfd_gini_lastbatch": None if np.isnan(apfd) else apfd,
+                        "apfd_dsa_lastbatch": None if np.isnan(apfd_dsa) else apfd_dsa,
+                        "apfd_pclsa_lastbatch": None if np.isnan(apfd_lsa) else apfd_lsa,
                     },
                 }
             ),
