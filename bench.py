@@ -46,20 +46,6 @@ def log(rank, msg):
         print(f"[bench] {msg}", file=sys.stderr, flush=True)
 
 
-@torch.no_grad()
-def extract_ats(model, x, device, batch=1024):
-    """Fused forward: AT tap + softmax in one pass (the K15 hot path)."""
-    ats, probs = [], []
-    use_amp = device.type == "cuda"
-    for s in range(0, x.shape[0], batch):
-        xb = x[s : s + batch].to(device, non_blocking=True)
-        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_amp):
-            taps, logits = model.forward_taps(xb, AT_TAP)
-        ats.append(taps[0].reshape(taps[0].shape[0], -1).float())
-        probs.append(torch.softmax(logits.float(), dim=1))
-    return torch.cat(ats), torch.cat(probs)
-
-
 class GraphedExtractor:
     """Inference-path AT extractor: BN-folded bf16 channels_last model,
     fixed-shape forward captured in a hipGraph (one replay per step).
