@@ -32,7 +32,15 @@ using bf16 = __hip_bfloat16;
 using short8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-TIP_DEV int swz(int u) { return u ^ ((u >> 4) & 7); }
+// 16-B-unit XOR swizzle family: u ^= ((u>>4) * M) & 15 (injective: only
+// bits 0-3 change, driven by bits >= 4). M chosen by an empirical sweep on
+// hardware (scripts/swz_sweep; the ds_read_b128 lane-service groups are
+// non-contiguous, so analytic conflict modelling is unreliable — see
+// profiles/r01_optimization_ladder.md).
+#ifndef TIP_SWZ_M
+#define TIP_SWZ_M 1
+#endif
+TIP_DEV int swz(int u) { return u ^ (((u >> 4) * TIP_SWZ_M) & 15); }
 
 // Load a 16-B unit (8 bf16) from the swizzled LDS image.
 TIP_DEV short8 lds_read_unit(const short* lds, int u) {
