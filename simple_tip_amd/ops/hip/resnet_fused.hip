@@ -33,14 +33,18 @@ using short8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 // 16-B-unit XOR swizzle family: u ^= ((u>>4) * M) & 15 (injective: only
-// bits 0-3 change, driven by bits >= 4). M chosen by an empirical sweep on
-// hardware (scripts/swz_sweep; the ds_read_b128 lane-service groups are
-// non-contiguous, so analytic conflict modelling is unreliable — see
-// profiles/r01_optimization_ladder.md).
+// bits 0-3 change, driven by bits >= 4). M = 0 (identity, default) won a
+// 9-variant hardware sweep by +22% (364 vs ~300 TF on resblock<32,32,16>):
+// the ds_read_b128 lane-service groups are non-contiguous, and for this
+// access pattern the linear image's residual conflicts cost less than the
+// per-access XOR address math (profiles/r01_optimization_ladder.md).
 #ifndef TIP_SWZ_M
-#define TIP_SWZ_M 1
+#define TIP_SWZ_M 0
 #endif
-TIP_DEV int swz(int u) { return u ^ (((u >> 4) * TIP_SWZ_M) & 15); }
+TIP_DEV int swz(int u) {
+  if (TIP_SWZ_M == 0) return u;
+  return u ^ (((u >> 4) * TIP_SWZ_M) & 15);
+}
 
 // Load a 16-B unit (8 bf16) from the swizzled LDS image.
 TIP_DEV short8 lds_read_unit(const short* lds, int u) {
