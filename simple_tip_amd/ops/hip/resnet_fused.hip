@@ -135,26 +135,27 @@ TIP_DEV void conv3x3(
       return (aoy * STRIDE) * RS + (aox * STRIDE) * UPP;
     };
     auto epilogue = [&](int pt, const f32x4& acc) {
+      // the 4 regs of a C/D fragment are 4 consecutive pixels in ONE output
+      // row (4 <= OW always), so the halo-image unit address is linear in
+      // reg: one div/mod per fragment, +OUPP per register
+      const int pix0 = pt * 16 + g * 4;
+      const int oy = pix0 / OW, ox = pix0 - oy * OW;
+      const int u0 = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
+      const int e = cout & 7;
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int pix = pt * 16 + g * 4 + reg;
         float v = acc[reg] + bs;
         if (RESID) {
-          const int oy = pix / OW, ox = pix - oy * OW;
-          const int ru = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
-          const short* runit = rlds + swz(ru) * 8;
-          const bf16 rv = reinterpret_cast<const bf16*>(runit)[cout & 7];
-          v += __bfloat162float(rv);
+          const short* runit = rlds + swz(u0 + reg * OUPP) * 8;
+          v += __bfloat162float(reinterpret_cast<const bf16*>(runit)[e]);
         }
         v = fmaxf(v, 0.f);
         const bf16 ov = __float2bfloat16(v);
         if (TO_LDS) {
-          const int oy = pix / OW, ox = pix - oy * OW;
-          const int ou = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
-          short* unit = out_lds + swz(ou) * 8;
-          reinterpret_cast<bf16*>(unit)[cout & 7] = ov;
+          short* unit = out_lds + swz(u0 + reg * OUPP) * 8;
+          reinterpret_cast<bf16*>(unit)[e] = ov;
         } else {
-          reinterpret_cast<bf16*>(gout)[(int64_t)pix * COUT + cout] = ov;
+          reinterpret_cast<bf16*>(gout)[(int64_t)(pix0 + reg) * COUT + cout] = ov;
         }
       }
     };
@@ -244,12 +245,12 @@ TIP_DEV void shortcut1x1_s2(
     }
     const int cout = ct * 16 + j;
     const float bs = bias[cout];
+    const int pix0 = p0 + g * 4;
+    const int oy = pix0 / OW, ox = pix0 - oy * OW;
+    const int u0 = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
-      const int pix = p0 + g * 4 + reg;
-      const int oy = pix / OW, ox = pix - oy * OW;
-      const int ou = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
-      short* unit = rlds + swz(ou) * 8;
+      short* unit = rlds + swz(u0 + reg * OUPP) * 8;
       reinterpret_cast<bf16*>(unit)[cout & 7] = __float2bfloat16(acc[reg] + bs);
     }
   }
