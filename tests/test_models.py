@@ -56,3 +56,42 @@ def test_forward_equals_forward_taps():
         direct = m(x)
     _, tapped = m.forward_taps(x, [0])
     assert torch.allclose(direct, tapped)
+
+
+def test_bn_folding_numerics():
+    from simple_tip_amd.models.fuse import fold_bn_inference
+
+    torch.manual_seed(3)
+    m = ResNet20()
+    for mod in m.modules():
+        if isinstance(mod, torch.nn.BatchNorm2d):
+            mod.running_mean.normal_(0, 0.3)
+            mod.running_var.uniform_(0.5, 2.0)
+            mod.weight.data.uniform_(0.5, 1.5)
+            mod.bias.data.normal_(0, 0.3)
+    m.eval()
+    folded = fold_bn_inference(m)
+    # no BatchNorm modules remain
+    assert not any(
+        isinstance(mod, torch.nn.BatchNorm2d) for mod in folded.modules()
+    )
+    x = torch.randn(4, 3, 32, 32)
+    with torch.no_grad():
+        a = m(x)
+        b = folded(x)
+    assert torch.allclose(a, b, rtol=1e-4, atol=1e-5)
+
+
+def test_pad_stem_channels_identity():
+    from simple_tip_amd.models.fuse import fold_bn_inference, pad_stem_channels
+
+    torch.manual_seed(4)
+    folded = fold_bn_inference(ResNet20()).eval()
+    x = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        ref = folded(x)
+    pad_stem_channels(folded, 4)
+    x4 = torch.cat([x, torch.zeros(2, 1, 32, 32)], dim=1)
+    with torch.no_grad():
+        out = folded(x4)
+    assert torch.allclose(ref, out, rtol=1e-5, atol=1e-6)
