@@ -130,6 +130,23 @@ class FusedPrioritizer:
                 )
                 self.lsa_ready = True
 
+    def _lsa_lse(self, padded, bp, tseg_cpu, tseg):
+        """Per-class whiten GEMMs + one grouped KDE launch."""
+        white = torch.zeros(bp, self.lsa_d, device=padded.device)
+        for c in range(self.num_classes):
+            lo, hi = int(tseg_cpu[c]), int(tseg_cpu[c + 1])
+            if hi <= lo or self.lsa_mode[c][0] != "fused":
+                continue
+            _, keep, linv_t = self.lsa_mode[c]
+            seg = padded[lo:hi]
+            if keep is not None:
+                seg = seg.index_select(1, keep)
+            white[lo:hi] = seg @ linv_t
+        return self.ext.grouped_kde(
+            white.contiguous(), self.lsa_wtrainS, tseg, self.lsa_nseg,
+            self.lsa_wnormS, self.lsa_jb_max,
+        )
+
     def _segment(self, ats: torch.Tensor, pred: torch.Tensor):
         """Class-sort + pad to 128-row segments."""
         counts = torch.bincount(pred, minlength=self.num_classes)
@@ -161,6 +178,20 @@ class FusedPrioritizer:
         padded = torch.zeros(bp, ats.shape[1], device=ats.device)
         padded[dest] = ats[order].float()
 
+        # DSA (main stream) and LSA whiten+KDE (side stream) are independent
+        # given `padded`; overlap them.
+        lse = None
+        main_stream = torch.cuda.current_stream() if ats.is_cuda else None
+        if self.lsa_ready and ats.is_cuda:
+            if not hasattr(self, "_lsa_stream"):
+                self._lsa_stream = torch.cuda.Stream()
+            ready = torch.cuda.Event()
+            ready.record(main_stream)
+            with torch.cuda.stream(self._lsa_stream):
+                self._lsa_stream.wait_event(ready)
+                lse = self._lsa_lse(padded, bp, tseg_cpu, tseg)
+                lse.record_stream(main_stream)
+
         dist, idx = self.ext.grouped_rowmin(
             padded.contiguous(), self.trainS, tseg, self.nseg, self.bnormS,
             self.jb_max,
@@ -175,20 +206,10 @@ class FusedPrioritizer:
 
         lsa = None
         if self.lsa_ready:
-            white = torch.zeros(bp, self.lsa_d, device=ats.device)
-            for c in range(self.num_classes):
-                lo, hi = int(tseg_cpu[c]), int(tseg_cpu[c + 1])
-                if hi <= lo or self.lsa_mode[c][0] != "fused":
-                    continue
-                _, keep, linv_t = self.lsa_mode[c]
-                seg = padded[lo:hi]
-                if keep is not None:
-                    seg = seg.index_select(1, keep)
-                white[lo:hi] = seg @ linv_t
-            lse = self.ext.grouped_kde(
-                white.contiguous(), self.lsa_wtrainS, tseg, self.lsa_nseg,
-                self.lsa_wnormS, self.lsa_jb_max,
-            )
+            if lse is None:  # CPU path (no side stream)
+                lse = self._lsa_lse(padded, bp, tseg_cpu, tseg)
+            else:
+                main_stream.wait_stream(self._lsa_stream)
             cls_of_row = torch.bucketize(
                 torch.arange(bp, device=ats.device), tseg.long()[1:], right=True
             )
