@@ -280,21 +280,29 @@ class MultiModalSA(SA):
 class MDSA(SA):
     """Mahalanobis-distance surprise adequacy (reference surprise.py:374-393).
 
-    Fit: float64 ML covariance + pseudo-inverse precision on the host.
-    Score: (x-mu) P (x-mu)^T row-dot — a GEMM + fused row-dot on device.
+    Fit: float64 ML covariance + pseudo-inverse precision — ON DEVICE when
+    the activations live there (K3: the D x D covariance is one fp64 GEMM,
+    the pseudo-inverse one rocSOLVER eigh; the CPU-sklearn fit was 45 s of
+    a full-scale run). Score: (x-mu) P (x-mu)^T row-dot GEMM.
     """
 
     def __init__(self, activations: Activations, device=None):
-        acts = _flatten_layers(activations).double().cpu()
-        self.mean = acts.mean(dim=0)
-        centered = acts - self.mean
+        acts_in = _flatten_layers(activations)
+        fit_dev = acts_in.device if acts_in.is_cuda else torch.device("cpu")
+        acts = acts_in.to(fit_dev, torch.float64)
+        mean = acts.mean(dim=0)
+        centered = acts - mean
         # sklearn EmpiricalCovariance: ML estimate (divide by N)
         cov = centered.t() @ centered / acts.shape[0]
-        self.precision = torch.linalg.pinv(cov, hermitian=True)
-        self.device = device
-        if device is not None and str(device) != "cpu":
-            self._mean_dev = self.mean.float().to(device)
-            self._prec_dev = self.precision.float().to(device)
+        precision = torch.linalg.pinv(cov, hermitian=True)
+        self.mean = mean.cpu()
+        self.precision = precision.cpu()
+        self.device = device if device is not None else (
+            fit_dev if fit_dev.type == "cuda" else None
+        )
+        if self.device is not None and str(self.device) != "cpu":
+            self._mean_dev = mean.float().to(self.device)
+            self._prec_dev = precision.float().to(self.device)
 
     def __call__(self, activations, predictions=None, num_threads=None):
         acts = _flatten_layers(activations)
@@ -406,20 +414,87 @@ class LSA(SA):
 
 class MLSA(SA):
     """Multimodal likelihood SA: -log GMM likelihood
-    (reference surprise.py:498-520). EM fit on the host (sklearn), scoring in
-    torch so it can run on device."""
+    (reference surprise.py:498-520).
+
+    CPU inputs fit with sklearn's EM (reference parity, the test oracle);
+    device inputs fit with a torch EM on the GPU (K4 — the CPU GMM fit was
+    84 s of a full-scale run, the device EM is seconds). Scoring is torch
+    either way. The two fits converge to (possibly) different local optima,
+    as any EM with different initialisation does.
+    """
+
+    REG_COVAR = 1e-6  # sklearn's default covariance regulariser
 
     def __init__(self, activations: Activations, num_components: int = 2, device=None):
+        acts_t = _flatten_layers(activations)
+        self.device = device
+        if acts_t.is_cuda:
+            self._fit_device(acts_t.double(), num_components)
+            return
         from sklearn.mixture import GaussianMixture
 
-        acts = _flatten_layers(activations).double().cpu().numpy()
+        acts = acts_t.double().cpu().numpy()
         self.gmm = GaussianMixture(n_components=num_components)
         self.gmm.fit(acts)
         self.means = torch.from_numpy(self.gmm.means_)  # [k, d]
         # precision cholesky P with Sigma^-1 = P P^T
         self.prec_chol = torch.from_numpy(self.gmm.precisions_cholesky_)
         self.log_weights = torch.from_numpy(np.log(self.gmm.weights_))
-        self.device = device
+
+    def _fit_device(self, x: torch.Tensor, k: int, iters: int = 60, tol: float = 1e-3):
+        """Full-covariance EM in fp64 on the device (kmeans init)."""
+        n, d = x.shape
+        centers, labels, _ = kmeans_fit(x.float(), k, n_init=1, max_iter=20, seed=0)
+        means = torch.stack(
+            [
+                x[labels == c].mean(dim=0) if bool((labels == c).any()) else x[c % n]
+                for c in range(k)
+            ]
+        )
+        weights = torch.full((k,), 1.0 / k, dtype=torch.float64, device=x.device)
+        covs = []
+        eye = torch.eye(d, dtype=torch.float64, device=x.device)
+        for c in range(k):
+            sel = labels == c
+            xc = (x[sel] if bool(sel.any()) else x) - means[c]
+            covs.append(xc.t() @ xc / max(int(sel.sum()), 1) + self.REG_COVAR * eye)
+        covs = torch.stack(covs)
+        prev_ll = None
+        for _ in range(iters):
+            # E-step: per-component log N(x; mu, Sigma) via Cholesky solves
+            log_prob = torch.empty(n, k, dtype=torch.float64, device=x.device)
+            chols = torch.linalg.cholesky(covs)
+            for c in range(k):
+                xc = (x - means[c]).t()
+                y = torch.linalg.solve_triangular(chols[c], xc, upper=False)
+                logdet = torch.log(torch.diagonal(chols[c])).sum()
+                log_prob[:, c] = (
+                    -0.5 * (y * y).sum(dim=0)
+                    - logdet
+                    - 0.5 * d * np.log(2 * np.pi)
+                    + torch.log(weights[c])
+                )
+            norm = torch.logsumexp(log_prob, dim=1)
+            ll = float(norm.mean())
+            resp = torch.exp(log_prob - norm.unsqueeze(1))
+            # M-step
+            nk = resp.sum(dim=0).clamp_min(1e-10)
+            means = (resp.t() @ x) / nk.unsqueeze(1)
+            for c in range(k):
+                xc = x - means[c]
+                covs[c] = (xc * resp[:, c : c + 1]).t() @ xc / nk[c] + self.REG_COVAR * eye
+            weights = nk / n
+            if prev_ll is not None and abs(ll - prev_ll) < tol:
+                break
+            prev_ll = ll
+        chols = torch.linalg.cholesky(covs)
+        # precision cholesky P = L^-T (Sigma^-1 = P P^T)
+        eye_b = eye.unsqueeze(0).expand(k, d, d)
+        self.prec_chol = torch.linalg.solve_triangular(
+            chols.transpose(1, 2), eye_b, upper=True
+        )
+        self.means = means
+        self.log_weights = torch.log(weights)
 
     def _score_samples(self, x: torch.Tensor) -> torch.Tensor:
         k, d = self.means.shape[0], self.means.shape[1]

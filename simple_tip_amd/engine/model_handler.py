@@ -157,10 +157,16 @@ class BaseModel:
                     # replicas per forward pass (throughput, not 200 passes)
                     reps = max(1, min(DROPOUT_SAMPLE_SIZE, 8192 // max(b, 1)))
                     done = 0
+                    use_amp = self.device.type == "cuda"
                     while done < DROPOUT_SAMPLE_SIZE:
                         r = min(reps, DROPOUT_SAMPLE_SIZE - done)
                         xrep = xb.repeat(r, *([1] * (xb.dim() - 1)))
-                        preds = self.model(xrep).argmax(dim=1).reshape(r, b)
+                        # bf16 autocast: the VR statistic is an argmax vote,
+                        # robust to reduced-precision logits
+                        with torch.autocast("cuda", dtype=torch.bfloat16,
+                                            enabled=use_amp):
+                            logits = self.model(xrep)
+                        preds = logits.argmax(dim=1).reshape(r, b)
                         votes.scatter_add_(
                             1,
                             preds.t(),
