@@ -247,7 +247,7 @@ def main():
     log(rank, "fitting DSA (0.3 subsample) + per-class LSA")
     dsa = DSA(train_ats, train_pred, subsampling=0.3, device=device)
     lsa = MultiModalSA.build_by_class(
-        train_ats.cpu(), train_pred.cpu(),
+        train_ats, train_pred,
         lambda a, p: LSA(a, max_features=300, device=device),
     )
     train_classes = set(torch.unique(train_pred.cpu()).tolist())

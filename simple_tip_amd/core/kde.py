@@ -96,7 +96,7 @@ class StableGaussianKDE:
 
     def whiten(self, points: torch.Tensor, dtype=None, device=None) -> torch.Tensor:
         """Whiten [m, d] points by the bandwidth Cholesky factor."""
-        pts = torch.as_tensor(points, dtype=torch.float64)
+        pts = torch.as_tensor(points, dtype=torch.float64).to(self.cho_l.device)
         y = torch.linalg.solve_triangular(self.cho_l, pts.t(), upper=False).t()
         y = y.contiguous()
         if dtype is not None or device is not None:
@@ -118,7 +118,7 @@ class StableGaussianKDE:
         if cache is None:
             cache = self._dev_cache = {}
         if key not in cache:
-            eye = torch.eye(self.d, dtype=torch.float64)
+            eye = torch.eye(self.d, dtype=torch.float64, device=self.cho_l.device)
             linv = torch.linalg.solve_triangular(self.cho_l, eye, upper=False)
             cache[key] = (
                 linv.t().to(device=device, dtype=dtype).contiguous(),

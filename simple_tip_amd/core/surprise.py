@@ -333,7 +333,9 @@ class LSA(SA):
         max_features: Optional[Union[int, float]] = 300,
         device=None,
     ):
-        acts = _flatten_layers(activations).double().cpu()
+        # fit stays on the activations' device (K2/K19: variance selection,
+        # covariance, jitter ladder and Cholesky are all device-capable)
+        acts = _flatten_layers(activations).double()
         assert var_threshold is None or max_features is None, (
             "var_threshold and max_features cannot both be specified"
         )
@@ -346,7 +348,7 @@ class LSA(SA):
                 num_features = int(min(max_features * acts.shape[1], acts.shape[1]))
             else:
                 num_features = int(min(max_features, acts.shape[1]))
-            var = acts.var(dim=0, unbiased=True).numpy()
+            var = acts.var(dim=0, unbiased=True).cpu().numpy()
             dropped = np.argsort(var, kind="stable")[:-num_features]
             self.removed_neurons = [int(x) for x in dropped]
         self.device = device
@@ -390,7 +392,9 @@ class LSA(SA):
     def _remove_unused_columns(self, acts: torch.Tensor) -> torch.Tensor:
         if self.removed_neurons:
             keep = np.delete(np.arange(acts.shape[1]), self.removed_neurons)
-            return acts[:, torch.from_numpy(keep)]
+            return acts.index_select(
+                1, torch.from_numpy(keep).to(acts.device)
+            )
         return acts
 
     def __call__(self, activations, predictions=None, num_threads=0):
