@@ -155,7 +155,7 @@ class BaseModel:
                     )
                     # vectorize the MC samples: R independent dropout
                     # replicas per forward pass (throughput, not 200 passes)
-                    reps = max(1, min(DROPOUT_SAMPLE_SIZE, 8192 // max(b, 1)))
+                    reps = max(1, min(DROPOUT_SAMPLE_SIZE, 32768 // max(b, 1)))
                     done = 0
                     use_amp = self.device.type == "cuda"
                     while done < DROPOUT_SAMPLE_SIZE:
