@@ -155,7 +155,9 @@ class BaseModel:
                     )
                     # vectorize the MC samples: R independent dropout
                     # replicas per forward pass (throughput, not 200 passes)
-                    reps = max(1, min(DROPOUT_SAMPLE_SIZE, 32768 // max(b, 1)))
+                    # macro-batch cap 8192: measured optimum (32768 made
+                    # MIOpen pick a 5x slower conv algo for these shapes)
+                    reps = max(1, min(DROPOUT_SAMPLE_SIZE, 8192 // max(b, 1)))
                     done = 0
                     use_amp = self.device.type == "cuda"
                     while done < DROPOUT_SAMPLE_SIZE:
