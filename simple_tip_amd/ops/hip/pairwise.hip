@@ -39,14 +39,11 @@ struct StageRegs {
   float4 v[4];
 };
 
-// BMT = the tile's row count; the block has BMT*2 threads, each staging 16
-// floats (one [BMT x BK] tile per call).
-template <int BMT>
 TIP_DEV StageRegs stage_load(
     const float* __restrict__ src, int rows, int K, int row0, int k0) {
   StageRegs sr;
   const int t = threadIdx.x;
-  const int r = t >> 1;                 // 0..BMT-1
+  const int r = t >> 1;                 // 0..127
   const int kq = (t & 1) * (BK / 2);    // 0 or 16
   const int grow = row0 + r;
   const bool row_ok = grow < rows;
@@ -69,17 +66,16 @@ TIP_DEV StageRegs stage_load(
   return sr;
 }
 
-template <int BMT>
 TIP_DEV void stage_write(float* lds, const StageRegs& sr) {
   const int t = threadIdx.x;
   const int r = t >> 1;
   const int kq = (t & 1) * (BK / 2);
 #pragma unroll
   for (int q4 = 0; q4 < 4; ++q4) {
-    lds[(kq + q4 * 4 + 0) * (BMT + LDS_PAD) + r] = sr.v[q4].x;
-    lds[(kq + q4 * 4 + 1) * (BMT + LDS_PAD) + r] = sr.v[q4].y;
-    lds[(kq + q4 * 4 + 2) * (BMT + LDS_PAD) + r] = sr.v[q4].z;
-    lds[(kq + q4 * 4 + 3) * (BMT + LDS_PAD) + r] = sr.v[q4].w;
+    lds[(kq + q4 * 4 + 0) * (BM + LDS_PAD) + r] = sr.v[q4].x;
+    lds[(kq + q4 * 4 + 1) * (BM + LDS_PAD) + r] = sr.v[q4].y;
+    lds[(kq + q4 * 4 + 2) * (BM + LDS_PAD) + r] = sr.v[q4].z;
+    lds[(kq + q4 * 4 + 3) * (BM + LDS_PAD) + r] = sr.v[q4].w;
   }
 }
 
@@ -128,17 +124,17 @@ __launch_bounds__(256, 2) __global__ void pairwise_kernel(
 
   // T14 pipeline: prologue stages tile 0; each iteration issues tile t+1's
   // global loads before tile t's MFMAs and writes them to LDS afterwards.
-  StageRegs ra = stage_load<BM>(A, M, K, row0, 0);
-  StageRegs rb = stage_load<BM>(B, N, K, col0, 0);
-  stage_write<BM>(As, ra);
-  stage_write<BM>(Bs, rb);
+  StageRegs ra = stage_load(A, M, K, row0, 0);
+  StageRegs rb = stage_load(B, N, K, col0, 0);
+  stage_write(As, ra);
+  stage_write(Bs, rb);
   __syncthreads();
 
   for (int k0 = 0; k0 < K; k0 += BK) {
     const bool has_next = (k0 + BK) < K;
     if (has_next) {
-      ra = stage_load<BM>(A, M, K, row0, k0 + BK);
-      rb = stage_load<BM>(B, N, K, col0, k0 + BK);
+      ra = stage_load(A, M, K, row0, k0 + BK);
+      rb = stage_load(B, N, K, col0, k0 + BK);
     }
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 2) {
@@ -154,8 +150,8 @@ __launch_bounds__(256, 2) __global__ void pairwise_kernel(
     }
     __syncthreads();  // all reads of tile t done
     if (has_next) {
-      stage_write<BM>(As, ra);
-      stage_write<BM>(Bs, rb);
+      stage_write(As, ra);
+      stage_write(Bs, rb);
     }
     __syncthreads();  // tile t+1 visible
   }
@@ -288,16 +284,16 @@ __launch_bounds__(256, 2) __global__ void grouped_pairwise_kernel(
   const int wc = wid & 1;
 
   f32x16 acc[2][2] = {};
-  StageRegs ra = stage_load<BM>(A, Bp, K, row0, 0);
-  StageRegs rb = stage_load<BM>(B, ncol1, K, col0, 0);
-  stage_write<BM>(As, ra);
-  stage_write<BM>(Bs, rb);
+  StageRegs ra = stage_load(A, Bp, K, row0, 0);
+  StageRegs rb = stage_load(B, ncol1, K, col0, 0);
+  stage_write(As, ra);
+  stage_write(Bs, rb);
   __syncthreads();
   for (int k0 = 0; k0 < K; k0 += BK) {
     const bool has_next = (k0 + BK) < K;
     if (has_next) {
-      ra = stage_load<BM>(A, Bp, K, row0, k0 + BK);
-      rb = stage_load<BM>(B, ncol1, K, col0, k0 + BK);
+      ra = stage_load(A, Bp, K, row0, k0 + BK);
+      rb = stage_load(B, ncol1, K, col0, k0 + BK);
     }
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 2) {
@@ -313,8 +309,8 @@ __launch_bounds__(256, 2) __global__ void grouped_pairwise_kernel(
     }
     __syncthreads();
     if (has_next) {
-      stage_write<BM>(As, ra);
-      stage_write<BM>(Bs, rb);
+      stage_write(As, ra);
+      stage_write(Bs, rb);
     }
     __syncthreads();
   }
@@ -360,136 +356,6 @@ __launch_bounds__(256, 2) __global__ void grouped_pairwise_kernel(
   }
   __syncthreads();
   for (int r = threadIdx.x; r < BM; r += blockDim.x) {
-    const int i = row0 + r;
-    if (i >= Bp) continue;
-    if (EPI == EPI_ROWMIN) {
-      MinIdx best = min_idx_combine(
-          MinIdx{red_v[r][0], red_i[r][0]}, MinIdx{red_v[r][1], red_i[r][1]});
-      pmin_val[(int64_t)bj * Bp + i] = best.v;
-      pmin_idx[(int64_t)bj * Bp + i] = best.i;
-    } else {
-      float m0 = red_v[r][0], s0 = red_s[r][0];
-      float m1 = red_v[r][1], s1 = red_s[r][1];
-      float mm, ss;
-      if (m0 >= m1) {
-        mm = m0;
-        ss = s0 + ((s1 > 0.f) ? s1 * __expf(m1 - m0) : 0.f);
-      } else {
-        mm = m1;
-        ss = s1 + ((s0 > 0.f) ? s0 * __expf(m0 - m1) : 0.f);
-      }
-      pkde[(int64_t)bj * Bp + i] = make_float2(mm, ss);
-    }
-  }
-}
-
-// 2-wave (BM=64 x BN=128) grouped variant: doubles the block count on the
-// narrow per-class segments where the 4-wave tile leaves the chip
-// under-occupied. Wave w owns columns w*64..w*64+64 of all 64 rows.
-template <int EPI>
-__launch_bounds__(128, 4) __global__ void grouped_pairwise_small_kernel(
-    const float* __restrict__ A, const float* __restrict__ B,
-    const float* __restrict__ anorm, const float* __restrict__ bnorm,
-    const int* __restrict__ tseg, const int* __restrict__ nseg, int nclasses,
-    int Bp, int K, float* __restrict__ pmin_val, int* __restrict__ pmin_idx,
-    float2* __restrict__ pkde) {
-  constexpr int BMS = 64;
-  __shared__ float As[BK * (BMS + LDS_PAD)];
-  __shared__ float Bs[BK * (BM + LDS_PAD)];  // B tile keeps 128 rows
-  __shared__ float red_v[BMS][2];
-  __shared__ int red_i[BMS][2];
-  __shared__ float red_s[BMS][2];
-
-  const int bi = blockIdx.y;
-  const int row0 = bi * BMS;
-  int cls = 0;
-  for (int c = 0; c < nclasses; ++c)
-    if (tseg[c] <= row0) cls = c;
-  const int ncol0 = nseg[cls], ncol1 = nseg[cls + 1];
-  const int col0 = ncol0 + blockIdx.x * BN;
-  if (col0 >= ncol1) return;
-  const int bj = blockIdx.x;
-
-  const int lane = lane_id();
-  const int wid = wave_id();  // 0..1
-  const int wc = wid;
-
-  f32x16 acc[2][2] = {};
-  // A: [64 x BK] via the narrow stager; B: one row per thread
-  StageRegs ra = stage_load<BMS>(A, Bp, K, row0, 0);
-  const int brow = col0 + threadIdx.x;
-  const bool brow_ok = brow < ncol1;
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    stage_write<BMS>(As, ra);
-    {
-      // B staging: thread t writes row t's BK values (k-major image)
-      const float* bsrc = B + (int64_t)brow * K + k0;
-#pragma unroll
-      for (int kk = 0; kk < BK; ++kk) {
-        float v = 0.f;
-        if (brow_ok && k0 + kk < K) v = bsrc[kk];
-        Bs[kk * (BM + LDS_PAD) + threadIdx.x] = v;
-      }
-    }
-    __syncthreads();
-    const bool has_next = (k0 + BK) < K;
-    if (has_next) ra = stage_load<BMS>(A, Bp, K, row0, k0 + BK);
-#pragma unroll
-    for (int kk = 0; kk < BK; kk += 2) {
-      const int k = kk + (lane >> 5);
-      const float* as = &As[k * (BMS + LDS_PAD) + (lane & 31)];
-      const float* bs = &Bs[k * (BM + LDS_PAD) + wc * 64 + (lane & 31)];
-      const float a0 = as[0], a1 = as[32];
-      const float b0 = bs[0], b1 = bs[32];
-      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
-    }
-    __syncthreads();
-  }
-
-  const int jl0 = col0 + wc * 64 + (lane & 31);
-  const float bn0 = (jl0 < ncol1) ? bnorm[jl0] : 0.f;
-  const float bn1 = (jl0 + 32 < ncol1) ? bnorm[jl0 + 32] : 0.f;
-#pragma unroll
-  for (int m = 0; m < 2; ++m) {
-#pragma unroll
-    for (int reg = 0; reg < 16; ++reg) {
-      const int row_local = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
-      const int block_row = m * 32 + row_local;
-      const int i = row0 + block_row;
-      const float an = (i < Bp) ? anorm[i] : 0.f;
-      float d0 = fmaxf(an + bn0 - 2.f * acc[m][0][reg], 0.f);
-      float d1 = fmaxf(an + bn1 - 2.f * acc[m][1][reg], 0.f);
-      if (jl0 >= ncol1) d0 = FLT_MAX;
-      if (jl0 + 32 >= ncol1) d1 = FLT_MAX;
-      if (EPI == EPI_ROWMIN) {
-        MinIdx mi{d0, jl0};
-        mi = min_idx_combine(mi, MinIdx{d1, jl0 + 32});
-        mi = half_reduce_min(mi);
-        if ((lane & 31) == 0) {
-          red_v[block_row][wc] = mi.v;
-          red_i[block_row][wc] = mi.i;
-        }
-      } else {
-        MinIdx mi{d0, 0};
-        mi = min_idx_combine(mi, MinIdx{d1, 0});
-        mi = half_reduce_min(mi);
-        const float tmax = -0.5f * mi.v;
-        float s = 0.f;
-        if (d0 != FLT_MAX) s += __expf(-0.5f * d0 - tmax);
-        if (d1 != FLT_MAX) s += __expf(-0.5f * d1 - tmax);
-        s = half_reduce_sum(s);
-        if ((lane & 31) == 0) {
-          red_v[block_row][wc] = tmax;
-          red_s[block_row][wc] = s;
-        }
-      }
-    }
-  }
-  __syncthreads();
-  for (int r = threadIdx.x; r < BMS; r += blockDim.x) {
     const int i = row0 + r;
     if (i >= Bp) continue;
     if (EPI == EPI_ROWMIN) {
@@ -652,8 +518,8 @@ void launch_grouped_rowmin(const float* a, const float* b, const float* an,
                            int nclasses, int bp, int k, int jb_max,
                            float* pval, int* pidx, float* out_dist,
                            int64_t* out_idx, hipStream_t s) {
-  dim3 grid(jb_max, ceil_div(bp, 64));
-  grouped_pairwise_small_kernel<EPI_ROWMIN><<<grid, 128, 0, s>>>(
+  dim3 grid(jb_max, ceil_div(bp, BM));
+  grouped_pairwise_kernel<EPI_ROWMIN><<<grid, 256, 0, s>>>(
       a, b, an, bn, tseg, nseg, nclasses, bp, k, pval, pidx, nullptr);
   grouped_rowmin_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
       pval, pidx, tseg, nseg, nclasses, bp, out_dist, out_idx);
@@ -663,8 +529,8 @@ void launch_grouped_kde(const float* a, const float* b, const float* an,
                         const float* bn, const int* tseg, const int* nseg,
                         int nclasses, int bp, int k, int jb_max, float2* pkde,
                         float* out_lse, hipStream_t s) {
-  dim3 grid(jb_max, ceil_div(bp, 64));
-  grouped_pairwise_small_kernel<EPI_KDE><<<grid, 128, 0, s>>>(
+  dim3 grid(jb_max, ceil_div(bp, BM));
+  grouped_pairwise_kernel<EPI_KDE><<<grid, 256, 0, s>>>(
       a, b, an, bn, tseg, nseg, nclasses, bp, k, nullptr, nullptr, pkde);
   grouped_kde_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
       pkde, tseg, nseg, nclasses, bp, out_lse);
