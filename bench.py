@@ -178,9 +178,20 @@ def main():
     ap.add_argument("--train-n", type=int, default=TRAIN_N)
     ap.add_argument("--setup-epochs", type=int, default=4,
                     help="untimed warm-up training epochs (class diversity)")
+    ap.add_argument("--shard-train", action="store_true",
+                    help="strong scaling: shard the train-AT axis across "
+                         "ranks; every rank scores the SAME batch against "
+                         "its shard, partials merge over RCCL")
     args = ap.parse_args()
 
     rank, world, device = pdist.init_from_env()
+    env_world = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.gpus != env_world:
+        raise SystemExit(
+            f"--gpus {args.gpus} but WORLD_SIZE={env_world}: for N>1 launch "
+            f"via `python -m torch.distributed.run --nnodes=1 "
+            f"--nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N`"
+        )
     on_gpu = device.type == "cuda"
     # note: torch.backends.cudnn.benchmark=True was measured WORSE here
     # (forward 16.4 -> 72.3 ms; MIOpen's "benchmark" find picked slower
@@ -228,15 +239,26 @@ def main():
     train_x = torch.from_numpy(base_x)
     extractor = GraphedExtractor(model, args.batch, device, use_graph=on_gpu)
 
-    log(rank, f"extracting train ATs ({args.train_n} x 4096)")
+    shard_train = bool(args.shard_train) and world > 1
+    if shard_train:
+        # each rank extracts ATs for its row shard, then the full train-AT
+        # tensor all-gathers (one-time, untimed setup) so SA fits are
+        # identical everywhere; DSA/LSA then keep only their row shards.
+        my_rows = train_x[pdist.shard_slice(args.train_n)]
+    else:
+        my_rows = train_x
+    log(rank, f"extracting train ATs ({my_rows.shape[0]} x 4096)")
     at_parts, pred_parts = [], []
-    for s in range(0, train_x.shape[0], 1024):
-        a, p = extractor(train_x[s : s + 1024])
+    for s in range(0, my_rows.shape[0], 1024):
+        a, p = extractor(my_rows[s : s + 1024])
         at_parts.append(a)
         pred_parts.append(p.argmax(dim=1))
     train_ats = torch.cat(at_parts)
     train_pred = torch.cat(pred_parts)
-    del train_x, at_parts, pred_parts
+    if shard_train:
+        train_ats = pdist.allgather_rows(train_ats, args.train_n)
+        train_pred = pdist.allgather_rows(train_pred, args.train_n)
+    del train_x, my_rows, at_parts, pred_parts
     if torch.unique(train_pred).numel() < 2:
         # last-resort guard (should not trigger after warm-up training):
         # use the synthetic labels so the per-class SA work is real
@@ -244,11 +266,16 @@ def main():
         train_pred = torch.from_numpy(base_y).to(train_pred.device)
 
     log(rank, f"train pred classes: {torch.bincount(train_pred.cpu(), minlength=10).tolist()}")
-    log(rank, "fitting DSA (0.3 subsample) + per-class LSA")
-    dsa = DSA(train_ats, train_pred, subsampling=0.3, device=device)
+    log(rank, "fitting DSA (0.3 subsample) + per-class LSA"
+              + (" [train axis sharded]" if shard_train else ""))
+    dsa = DSA(
+        train_ats, train_pred, subsampling=0.3, device=device,
+        shard_train=shard_train,
+    )
     lsa = MultiModalSA.build_by_class(
         train_ats, train_pred,
-        lambda a, p: LSA(a, max_features=300, device=device),
+        lambda a, p: LSA(a, max_features=300, device=device,
+                         shard_train=shard_train),
     )
     train_classes = set(torch.unique(train_pred.cpu()).tolist())
 
@@ -272,8 +299,11 @@ def main():
     nhwc_pool = extractor.fused is not None
     from simple_tip_amd.studies.synthetic import corrupt_images
     for i in range(n_pool):
+        # shard-train mode: every rank scores the SAME global batch (strong
+        # scaling), so the pool seed must not depend on rank
+        tag = f"test-shared-{i}" if shard_train else f"test-r{rank}-{i}"
         px, py = synthetic_images(
-            "bench_cifar10", f"test-r{rank}-{i}", args.batch, (3, 32, 32), 10
+            "bench_cifar10", tag, args.batch, (3, 32, 32), 10
         )
         # reference OOD recipe: half the batch is corrupted, so the model
         # has real faults for the APFD quality signal
@@ -356,8 +386,10 @@ def main():
             lsa_scores = lsa(ats, pred)
             mark("lsa")
         gini = unc["deep_gini"]
-        # publish score shards (tiny, latency-bound on xGMI)
-        if world > 1:
+        # publish score shards (tiny, latency-bound on xGMI). In shard-train
+        # mode the DSA/LSA partials already merged inside the prioritizer and
+        # every rank holds full-batch scores — nothing left to gather.
+        if world > 1 and not shard_train:
             n_total = args.batch * world
             gini_all = pdist.allgather_rows(gini, n_total)
             dsa_all = pdist.allgather_rows(dsa_scores.float().to(device), n_total)
@@ -388,7 +420,8 @@ def main():
     elapsed = time.perf_counter() - t0
     elapsed = pdist.allreduce_max_scalar(elapsed, device)
 
-    total_inputs = args.batch * world * args.steps
+    per_step_inputs = args.batch if shard_train else args.batch * world
+    total_inputs = per_step_inputs * args.steps
     value = total_inputs / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
@@ -423,15 +456,17 @@ def main():
                     "warmup": args.warmup,
                     "ms_per_step": ms_per_step,
                     "higher_is_better": True,
-                    "scaling": "weak",
+                    "scaling": "strong" if shard_train else "weak",
                     "vs_baseline": None,
                     "dtype": "bf16",
                     "data": "synthetic",
                     "config": {
                         "model": "cifar10_resnet20",
-                        "global_batch": args.batch * world,
+                        "global_batch": per_step_inputs,
                         "seq_len": None,
-                        "parallelism": f"dp{world}",
+                        "parallelism": (
+                            f"train-shard{world}" if shard_train else f"dp{world}"
+                        ),
                         "train_ats": args.train_n,
                         "at_width": 4096,
                         "scorers": "gini+softmax-family+dsa+pc-lsa",

@@ -127,17 +127,36 @@ class StableGaussianKDE:
             )
         return cache[key]
 
-    def log_density_device(self, points: torch.Tensor) -> torch.Tensor:
-        """Device-resident log pdf for fp32 points already on the GPU."""
+    def log_density_device(
+        self, points: torch.Tensor, shard_train: bool = False
+    ) -> torch.Tensor:
+        """Device-resident log pdf for fp32 points already on the GPU.
+
+        ``shard_train`` evaluates the kernel sum over this rank's row-shard
+        of the whitened training set and merges the partial logsumexps
+        across ranks (parallel/sharded.py) — the normalising constant uses
+        the GLOBAL n, so the merged result equals the unsharded one.
+        """
         if self.prepare_failed:
             return torch.full(
                 (points.shape[0],), float("-inf"), device=points.device
             )
         linv_t, xw, const = self.device_state(points.device, points.dtype)
         y = (points @ linv_t).contiguous()
-        return ops.kde_logsumexp(y, xw) + const
+        return self._lse(y, xw, shard_train) + const
 
-    def log_density(self, points: torch.Tensor, device=None) -> torch.Tensor:
+    @staticmethod
+    def _lse(tw: torch.Tensor, xw: torch.Tensor, shard_train: bool) -> torch.Tensor:
+        if shard_train:
+            from ..parallel import sharded as shd
+
+            xw_local, _ = shd.shard_rows(xw)
+            return shd.sharded_kde_logsumexp(tw, xw_local)
+        return ops.kde_logsumexp(tw, xw)
+
+    def log_density(
+        self, points: torch.Tensor, device=None, shard_train: bool = False
+    ) -> torch.Tensor:
         """log pdf at [m, d] points.
 
         On a GPU device the pairwise kernel runs in fp32 on the MFMA path
@@ -149,11 +168,11 @@ class StableGaussianKDE:
         if device is not None and str(device) != "cpu":
             tw = self.whiten(points, dtype=torch.float32, device=device)
             xw = self.white_train(dtype=torch.float32, device=device)
-            lse = ops.kde_logsumexp(tw, xw).double().cpu()
+            lse = self._lse(tw, xw, shard_train).double().cpu()
         else:
             tw = self.whiten(points)
             xw = self.white_train()
-            lse = ops.kde_logsumexp(tw, xw)
+            lse = self._lse(tw, xw, shard_train)
         return lse - np.log(self.n) - 0.5 * self.log_det
 
     def evaluate(self, points: torch.Tensor) -> torch.Tensor:

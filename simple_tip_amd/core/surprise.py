@@ -11,6 +11,7 @@ operands to the device.
 """
 
 import abc
+import os
 import warnings
 from typing import Callable, Dict, Iterable, List, Optional, Tuple, Union
 
@@ -25,6 +26,20 @@ from .kmeans import kmeans_fit, kmeans_predict, silhouette_score
 Activations = Union[List, np.ndarray, torch.Tensor]
 Predictions = Union[List, np.ndarray, torch.Tensor]
 Discriminator = Callable[[torch.Tensor, Optional[torch.Tensor]], torch.Tensor]
+
+# Strict mode: inputs routed to a mode/class unseen at fit time RAISE (the
+# reference's behaviour, surprise.py:308-315) instead of scoring +inf with a
+# warning. Off by default; parity-validation runs flip it on to detect the
+# divergence instead of absorbing it.
+_STRICT_MODES = os.environ.get("TIP_STRICT_MODES") == "1"
+
+
+def set_strict_modes(flag: bool) -> bool:
+    """Toggle strict unseen-mode handling; returns the previous value."""
+    global _STRICT_MODES
+    prev = _STRICT_MODES
+    _STRICT_MODES = bool(flag)
+    return prev
 
 
 # ---------------------------------------------------------------------------
@@ -257,7 +272,13 @@ class MultiModalSA(SA):
                 # data (e.g. a predicted class absent from the training
                 # predictions). The reference raises here
                 # (surprise.py:308-315); we treat such inputs as maximally
-                # surprising instead, consistent with DSA's handling.
+                # surprising instead, consistent with DSA's handling —
+                # unless strict mode is on (set_strict_modes / TIP_STRICT_MODES).
+                if _STRICT_MODES:
+                    raise ValueError(
+                        f"No modal found for modal id {modal_id} "
+                        f"(reference raises here)"
+                    )
                 warnings.warn(
                     f"No modal found for modal id {modal_id}; scoring those "
                     f"inputs as maximally surprising (+inf).",
@@ -332,9 +353,17 @@ class LSA(SA):
         var_threshold: Optional[float] = None,
         max_features: Optional[Union[int, float]] = 300,
         device=None,
+        shard_train: bool = False,
     ):
         # fit stays on the activations' device (K2/K19: variance selection,
         # covariance, jitter ladder and Cholesky are all device-capable)
+        from ..parallel.dist import get_world_size
+
+        # train-axis sharding: the KDE fit happens on the full
+        # (replicated) class data — identical on every rank — but evaluation
+        # sums the kernel over this rank's whitened-train row shard only,
+        # merging partial logsumexps across ranks.
+        self.shard_train = bool(shard_train) and get_world_size() > 1
         acts = _flatten_layers(activations).double()
         assert var_threshold is None or max_features is None, (
             "var_threshold and max_features cannot both be specified"
@@ -407,12 +436,14 @@ class LSA(SA):
                     np.delete(np.arange(acts.shape[1]), self.removed_neurons)
                 ).to(acts.device)
                 acts = acts.index_select(1, keep)
-            return -self.kde.log_density_device(acts.float().contiguous())
+            return -self.kde.log_density_device(
+                acts.float().contiguous(), shard_train=self.shard_train
+            )
         dev = acts.device if acts.is_cuda else self.device
         acts = self._remove_unused_columns(acts.double().cpu())
         if self.kde is None:
             return torch.zeros(acts.shape[0], dtype=torch.float64)
-        logd = self.kde.log_density(acts, device=dev)
+        logd = self.kde.log_density(acts, device=dev, shard_train=self.shard_train)
         return -logd
 
 
@@ -434,6 +465,7 @@ class MLSA(SA):
         self.device = device
         if acts_t.is_cuda:
             self._fit_device(acts_t.double(), num_components)
+            self._sync_fit()
             return
         from sklearn.mixture import GaussianMixture
 
@@ -444,6 +476,22 @@ class MLSA(SA):
         # precision cholesky P with Sigma^-1 = P P^T
         self.prec_chol = torch.from_numpy(self.gmm.precisions_cholesky_)
         self.log_weights = torch.from_numpy(np.log(self.gmm.weights_))
+        self._sync_fit()
+
+    def _sync_fit(self):
+        """Broadcast the fit from rank 0 so every rank scores with identical
+        parameters: EM init is the one nondeterministic fit in the SA family
+        (sklearn seeds from the process RNG; the device EM's kmeans init is
+        seeded but EM converges to a local optimum either way). Part of the
+        "broadcast statistics at setup" collective set."""
+        from ..parallel.dist import is_initialized
+
+        if not is_initialized():
+            return
+        import torch.distributed as dist
+
+        for t in (self.means, self.prec_chol, self.log_weights):
+            dist.broadcast(t, src=0)
 
     def _fit_device(self, x: torch.Tensor, k: int, iters: int = 60, tol: float = 1e-3):
         """Full-covariance EM in fp64 on the device (kmeans init)."""
@@ -533,6 +581,15 @@ class DSA(SA):
     The reference slices work into 10-row "badges" on a 5-thread pool to
     bound numpy's N x M x D broadcast; here each class is one fused
     pairwise-distance + row-min/argmin kernel launch on the device.
+
+    ``shard_train=True`` shards the train-AT axis across the ranks of an
+    initialised torch.distributed group: every rank keeps
+    only its contiguous row-shard of each class, scores the (replicated)
+    inputs against the shard, and the per-rank partial (min, argmin) merge
+    with a deterministic rank-ordered reduction, so 1-GPU and N-GPU scores
+    are identical. The two-hop b-table is fit sharded too (each rank
+    computes the rows of its shard) and then all-gathered, since scoring
+    needs b-table entries of whichever global row wins the merge.
     """
 
     def __init__(
@@ -543,6 +600,7 @@ class DSA(SA):
         subsampling: Union[int, float] = 1.0,
         subsampling_seed: int = 0,
         device=None,
+        shard_train: bool = False,
     ):
         acts = _flatten_layers(activations)
         preds = _class_predictions(predictions)
@@ -555,37 +613,72 @@ class DSA(SA):
         self.train_predictions = preds
         self.num_classes = int(preds.max().item()) + 1
         self.badge_size = badge_size  # kept for API parity; kernels batch freely
+        from ..parallel.dist import get_world_size
+
+        self.shard_train = bool(shard_train) and get_world_size() > 1
         self._class_cache = None  # device path: per-class (same_ats, b_table)
 
     def _build_class_cache(self):
-        """Per class c: (same-class ATs contiguous, b_table) where
-        b_table[i] = distance from same-class AT i to its nearest
-        OTHER-class training AT.
+        """Per class c: (same-class ATs contiguous, b_table, same_norm,
+        shard_offset) where b_table[i] = distance from same-class AT i to its
+        nearest OTHER-class training AT.
 
         dist_b depends only on which training AT is closest to the input
         (two-hop), so it is a fixed function of the training set: we
         precompute it once at fit time (part of the SA "setup" timing
         bucket) and the per-input path becomes hop-1 + a gather.
+
+        With ``shard_train`` the stored same-class ATs are this rank's row
+        shard (offset recorded) while the b-table stays global: the b-table
+        rows of the local shard are computed here (1/world of the fit work)
+        and all-gathered.
         """
+        from ..parallel import sharded as shd
+        from ..parallel.dist import allgather_rows
+
         cache = {}
         for label in range(self.num_classes):
             same_sel = self.train_predictions == label
             same = self.train_activations[same_sel].contiguous()
             if same.shape[0] == 0:
-                cache[label] = (None, None, None)
+                cache[label] = (None, None, None, 0)
                 continue
-            same_norm = (
-                (same.float() * same.float()).sum(dim=1).contiguous()
-                if same.is_cuda
-                else None
-            )
             other = self.train_activations[~same_sel].contiguous()
+            if self.shard_train:
+                n_same = same.shape[0]
+                local, off = shd.shard_rows(same)
+                if other.shape[0] == 0:
+                    # single-class training set: an all-inf b-table keeps the
+                    # global indexing consistent across ranks and yields
+                    # dsa = dist/inf = 0 ("no contrast"), like the dense path
+                    b_table = torch.full(
+                        (n_same,), float("inf"),
+                        dtype=same.dtype, device=same.device,
+                    )
+                    cache[label] = (local, b_table, self._rownorm(local), off)
+                    continue
+                if local.shape[0] > 0:
+                    bt_local, _ = ops.rowmin_l2(local, other)
+                else:
+                    bt_local = torch.empty(
+                        0, dtype=same.dtype, device=same.device
+                    )
+                b_table = allgather_rows(bt_local, n_same)
+                cache[label] = (local, b_table, self._rownorm(local), off)
+                continue
             if other.shape[0] == 0:
-                cache[label] = (same, None, same_norm)
+                cache[label] = (same, None, self._rownorm(same), 0)
                 continue
             b_table, _ = ops.rowmin_l2(same, other)
-            cache[label] = (same, b_table, same_norm)
+            cache[label] = (same, b_table, self._rownorm(same), 0)
         self._class_cache = cache
+
+    @staticmethod
+    def _rownorm(t: torch.Tensor):
+        """Cached ||row||^2 for the GPU kernels (None on CPU/empty)."""
+        if not t.is_cuda or t.shape[0] == 0:
+            return None
+        return (t.float() * t.float()).sum(dim=1).contiguous()
 
     def __call__(self, activations, predictions, num_threads=None):
         target_ats = _flatten_layers(activations)
@@ -609,20 +702,32 @@ class DSA(SA):
             if counts[label]:
                 # predicted class never seen in training predictions:
                 # maximally surprising (the reference would crash here)
+                if _STRICT_MODES:
+                    raise ValueError(
+                        f"Predicted class {label} was never predicted on the "
+                        f"training set (reference raises here)"
+                    )
                 dsa[target_pred == label] = float("inf")
+        if self.shard_train:
+            from ..parallel import sharded as shd
         for label in range(self.num_classes):
             if counts[label] == 0:
                 continue
             sel = target_pred == label
-            same, b_table, same_norm = self._class_cache[label]
+            same, b_table, same_norm, off = self._class_cache[label]
             if same is None:
                 dsa[sel] = float("inf")
                 continue
             samples = target_ats[sel]
-            dist_a, closest_idx = ops.rowmin_l2(samples, same, same_norm)
             if b_table is None:
                 dsa[sel] = 0.0  # single-class training set: no contrast
                 continue
+            if self.shard_train:
+                dist_a, closest_idx = shd.sharded_rowmin_l2(
+                    samples, same, off, same_norm
+                )
+            else:
+                dist_a, closest_idx = ops.rowmin_l2(samples, same, same_norm)
             dsa[sel] = dist_a / b_table[closest_idx]
         return dsa
 

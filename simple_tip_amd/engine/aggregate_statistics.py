@@ -98,6 +98,28 @@ class AggregateStatisticsCollector:
             for i, layer in enumerate(flat):
                 self.welfords[i].add_batch(layer)
 
+    def cross_rank_merge(self) -> None:
+        """Combine per-rank partial statistics across an initialised
+        torch.distributed group (K18 cross-GPU combine): elementwise min/max
+        all-reduce plus a rank-ordered Chan merge of the Welford partials.
+        Call once, after tracking this rank's train shard, before `get`."""
+        from ..parallel import sharded as shd
+        from ..parallel.dist import is_initialized
+
+        if not is_initialized():
+            return
+        with self.min_timer:
+            for m in self.mins:
+                shd.allreduce_min(m)
+        with self.max_timer:
+            for m in self.maxs:
+                shd.allreduce_max(m)
+        with self.welford_timer:
+            for w in self.welfords:
+                w.count, w.mean, w.m2 = shd.allreduce_welford(
+                    w.count, w.mean, w.m2
+                )
+
     def get(self) -> AggStats:
         """(mins, maxs, stds) per layer (flattened per-neuron vectors)."""
         self.done = True

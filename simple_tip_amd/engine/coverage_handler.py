@@ -23,24 +23,41 @@ logger = logging.getLogger(__name__)
 
 
 class CoverageWorker:
-    """Fits aggregate train statistics once; evaluates all NC metrics."""
+    """Fits aggregate train statistics once; evaluates all NC metrics.
 
-    def __init__(self, base_model: BaseModel, training_set):
+    ``dist_shard=True`` on an initialised torch.distributed group shards
+    BOTH passes across ranks: the train-statistics pass walks this rank's
+    row shard and Chan/min/max-merges the partials (K18 collectives), and
+    ``evaluate_all`` profiles this rank's test shard, reassembling the full
+    profile matrix with the coverage-bitmap OR all-reduce (BASELINE
+    config 4) so scores and the CAM order are identical on every rank.
+    """
+
+    def __init__(self, base_model: BaseModel, training_set, dist_shard: bool = False):
+        from ..parallel.dist import get_world_size, shard_slice
         from .aggregate_statistics import AggregateStatisticsCollector
 
         self.base_model = base_model
+        self.dist_shard = bool(dist_shard) and get_world_size() > 1
         self.metrics: Dict[str, CoverageMethod] = {}
         self.setup_times: Dict[str, float] = {}
 
+        train_part = (
+            training_set[shard_slice(training_set.shape[0])]
+            if self.dist_shard
+            else training_set
+        )
         agg = AggregateStatisticsCollector()
         pred_timer = DeviceTimer(start=True)
         for acts in base_model.walk_activations(
-            iter_batches(training_set, base_model.predict_batch)
+            iter_batches(train_part, base_model.predict_batch)
         ):
             pred_timer.stop()
             agg.track(acts)
             pred_timer.start()
         pred_timer.stop()
+        if self.dist_shard:
+            agg.cross_rank_merge()
         mins, maxs, stds = agg.get()
 
         nbc_debit = (
@@ -94,8 +111,16 @@ class CoverageWorker:
         scores_parts: Dict[str, List[torch.Tensor]] = {m: [] for m in self.metrics}
         profile_parts: Dict[str, List[BitProfile]] = {m: [] for m in self.metrics}
 
+        n_total = test_dataset.shape[0]
+        if self.dist_shard:
+            from ..parallel.dist import shard_slice
+
+            my_slice = shard_slice(n_total)
+            test_part = test_dataset[my_slice]
+        else:
+            test_part = test_dataset
         gen = self.base_model.walk_activations(
-            iter_batches(test_dataset, self.base_model.predict_batch)
+            iter_batches(test_part, self.base_model.predict_batch)
         )
         while True:
             t = DeviceTimer()
@@ -119,6 +144,10 @@ class CoverageWorker:
         for metric_id in self.metrics.keys():
             scores = torch.cat(scores_parts[metric_id])
             profile = BitProfile.cat(profile_parts[metric_id])
+            if self.dist_shard:
+                scores, profile = self._assemble_shards(
+                    scores, profile, n_total, my_slice
+                )
             all_scores[metric_id] = scores.cpu().numpy()
             logger.info("Calculating CAM for %s (%s)", metric_id, test_dataset_id)
             timer = DeviceTimer()
@@ -128,6 +157,28 @@ class CoverageWorker:
             self._cam_sanity_check(order, all_scores[metric_id])
             cam_orders[metric_id] = order
         return times, all_scores, cam_orders
+
+    @staticmethod
+    def _assemble_shards(scores, profile: BitProfile, n_total: int, my_slice):
+        """Reassemble full (scores, profiles) from per-rank input shards.
+
+        The profile rows land in a zero-initialised full-size word matrix and
+        a bitwise-OR all-reduce unions them — rows are disjoint across ranks,
+        so the OR is exact and every rank ends with the identical full
+        matrix for CAM. Scores all-gather (reference semantics
+        handler_coverage.py:189-205, sharded per SURVEY §2.4).
+        """
+        from ..parallel.dist import allgather_rows
+        from ..parallel.sharded import allreduce_bitmap_or
+
+        full_words = torch.zeros(
+            n_total, profile.words.shape[1],
+            dtype=profile.words.dtype, device=profile.words.device,
+        )
+        full_words[my_slice] = profile.words
+        allreduce_bitmap_or(full_words)
+        full_scores = allgather_rows(scores, n_total)
+        return full_scores, BitProfile(full_words, profile.nbits)
 
     @staticmethod
     def _cam_sanity_check(cam_order, scores):

@@ -19,7 +19,15 @@ from .surprise_handler import SurpriseHandler
 logger = logging.getLogger(__name__)
 
 
+def _is_rank0() -> bool:
+    from ..parallel.dist import get_rank
+
+    return get_rank() == 0
+
+
 def _persist(case_study: str, dataset_id: str, data_type: str, model_id: int, data):
+    if not _is_rank0():
+        return  # ranks hold identical artifacts; one writer avoids races
     ensure_dir("priorities")
     np.save(
         assets_path(
@@ -30,6 +38,8 @@ def _persist(case_study: str, dataset_id: str, data_type: str, model_id: int, da
 
 
 def _persist_times(case_study: str, dataset_id: str, model_id: int, metric: str, data: List[float]):
+    if not _is_rank0():
+        return
     ensure_dir("times")
     with open(
         assets_path("times", f"{case_study}_{dataset_id}_{model_id}_{metric}"), "wb"
@@ -66,30 +76,59 @@ def evaluate(
     dsa_badge_size: Optional[int] = None,
     device=None,
     predict_batch: int = 512,
+    dist_shard: bool = False,
 ) -> None:
-    """Run all TIP families for one model and persist every artifact."""
+    """Run all TIP families for one model and persist every artifact.
+
+    ``dist_shard=True`` on an initialised torch.distributed group runs the
+    whole experiment data-parallel over the test-input axis (RCCL on GPU,
+    gloo on CPU): every rank forwards/scores its contiguous input shard,
+    per-input scores all-gather, coverage profiles reassemble with the
+    bitmap OR all-reduce, and CAM runs identically on every rank from the
+    full matrices. Artifacts are written by rank 0 and are identical to a
+    single-process run (same batch content per forward).
+    """
     _eval_fault_predictors(
         case_study, model, model_id, nominal_test_dataset, nominal_test_labels,
-        "nominal", device, predict_batch,
+        "nominal", device, predict_batch, dist_shard,
     )
     _eval_fault_predictors(
         case_study, model, model_id, ood_test_dataset, ood_test_labels, "ood",
-        device, predict_batch,
+        device, predict_batch, dist_shard,
     )
     _eval_neuron_coverage(
         case_study, model, model_id, nc_activation_layers,
-        nominal_test_dataset, ood_test_dataset, training_dataset, device, predict_batch,
+        nominal_test_dataset, ood_test_dataset, training_dataset, device,
+        predict_batch, dist_shard,
     )
     _eval_surprise(
         case_study, model, model_id, sa_activation_layers,
         nominal_test_dataset, ood_test_dataset, training_dataset,
-        dsa_badge_size, device, predict_batch,
+        dsa_badge_size, device, predict_batch, dist_shard,
     )
 
 
-def _eval_fault_predictors(case_study, model, model_id, ds, labels, ds_type, device, predict_batch):
+def _eval_fault_predictors(
+    case_study, model, model_id, ds, labels, ds_type, device, predict_batch,
+    dist_shard=False,
+):
+    from ..parallel.dist import allgather_rows, get_world_size, shard_slice
+
     base_model = BaseModel(model, activation_layers=None, device=device, predict_batch=predict_batch)
-    pred, uncertainties, times = base_model.get_pred_and_uncertainty(ds)
+    if dist_shard and get_world_size() > 1:
+        import torch
+
+        n = ds.shape[0]
+        pred_l, unc_l, times = base_model.get_pred_and_uncertainty(
+            ds[shard_slice(n)]
+        )
+        pred = allgather_rows(torch.from_numpy(pred_l), n).numpy()
+        uncertainties = {
+            k: allgather_rows(torch.from_numpy(v), n).numpy()
+            for k, v in unc_l.items()
+        }
+    else:
+        pred, uncertainties, times = base_model.get_pred_and_uncertainty(ds)
     is_misclassified = pred != np.asarray(labels).reshape(-1)
     _persist(case_study, ds_type, "is_misclassified", model_id, is_misclassified)
     _persist_times_multiple_metrics(case_study, ds_type, model_id, times)
@@ -99,11 +138,12 @@ def _eval_fault_predictors(case_study, model, model_id, ds, labels, ds_type, dev
 
 def _eval_neuron_coverage(
     case_study, model, model_id, layers, nominal_test_dataset, ood_test_dataset,
-    training_dataset, device, predict_batch,
+    training_dataset, device, predict_batch, dist_shard=False,
 ):
     nc_worker = CoverageWorker(
         base_model=BaseModel(model, activation_layers=layers, device=device, predict_batch=predict_batch),
         training_set=training_dataset,
+        dist_shard=dist_shard,
     )
     for name, ds in {"nominal": nominal_test_dataset, "ood": ood_test_dataset}.items():
         times, scores, cam_orders = nc_worker.evaluate_all(ds, name)
@@ -116,11 +156,11 @@ def _eval_neuron_coverage(
 
 def _eval_surprise(
     case_study, model, model_id, layers, nominal_test_dataset, ood_test_dataset,
-    training_dataset, dsa_badge_size, device, predict_batch,
+    training_dataset, dsa_badge_size, device, predict_batch, dist_shard=False,
 ):
     sa_worker = SurpriseHandler(
         model=model, sa_layers=layers, training_dataset=training_dataset,
-        device=device, predict_batch=predict_batch,
+        device=device, predict_batch=predict_batch, dist_shard=dist_shard,
     )
     results = sa_worker.evaluate_all(
         datasets={"nominal": nominal_test_dataset, "ood": ood_test_dataset},

@@ -23,7 +23,16 @@ from ..ops import _load_compiled
 
 
 class FusedPrioritizer:
-    """Batched DSA + per-class LSA scorer on the grouped kernels (GPU)."""
+    """Batched DSA + per-class LSA scorer on the grouped kernels (GPU).
+
+    When the DSA (and the LSA instances) were fit with ``shard_train`` on an
+    initialised torch.distributed group, the train side of the grouped
+    kernels is this rank's per-class row shard: every rank scores the SAME
+    (replicated) batch against 1/world of the train rows, then the partial
+    (min, argmin) / logsumexp results merge with deterministic rank-ordered
+    reductions over RCCL (parallel/sharded.py semantics), so sharded scores
+    equal single-device ones.
+    """
 
     def __init__(self, dsa: DSA, lsa: Optional[MultiModalSA], device):
         self.ext = _load_compiled()
@@ -31,26 +40,44 @@ class FusedPrioritizer:
         if dsa._class_cache is None:
             dsa._build_class_cache()
         self.num_classes = dsa.num_classes
+        self.shard_train = bool(getattr(dsa, "shard_train", False))
 
-        # class-concatenated DSA train side + global b-table
+        # class-concatenated DSA train side + GLOBAL b-table. In shard mode
+        # the cache's `same` is the local row shard while `b_table` is the
+        # full (all-gathered) per-class table, so `btableS` is identical on
+        # every rank and the local→global row map sends grouped-kernel argmin
+        # indices (local concatenation) to full-concatenation rows.
         trains, btables, offs = [], [], [0]
+        l2g_parts = []
+        g_off = 0
         for c in range(self.num_classes):
-            same, b_table, _ = dsa._class_cache[c]
+            same, b_table, _, off = dsa._class_cache[c]
             if same is None:
                 offs.append(offs[-1])
                 continue
+            n_full = b_table.shape[0] if b_table is not None else same.shape[0]
             trains.append(same.float())
             bt = (
-                b_table.float()
+                b_table.float().to(device)
                 if b_table is not None
-                else torch.full((same.shape[0],), float("inf"), device=device)
+                else torch.full((n_full,), float("inf"), device=device)
             )
             btables.append(bt)
             offs.append(offs[-1] + same.shape[0])
+            if self.shard_train:
+                l2g_parts.append(
+                    g_off + off + torch.arange(
+                        same.shape[0], dtype=torch.int64, device=device
+                    )
+                )
+            g_off += n_full
         self.trainS = torch.cat(trains).contiguous().to(device)
         self.btableS = torch.cat(btables).contiguous().to(device)
         self.bnormS = (self.trainS * self.trainS).sum(dim=1).contiguous()
         self.nseg = torch.tensor(offs, dtype=torch.int32, device=device)
+        self.local2global = (
+            torch.cat(l2g_parts) if self.shard_train else None
+        )
         self.jb_max = max(
             1,
             max(
@@ -88,6 +115,10 @@ class FusedPrioritizer:
                     woffs.append(woffs[-1])
                     continue
                 linv_t, xw, const = sa.kde.device_state(device)
+                if self.shard_train:
+                    from ..parallel.sharded import shard_rows
+
+                    xw, _ = shard_rows(xw)
                 if d_ref is None:
                     d_ref = linv_t.shape[0]
                 if linv_t.shape[0] != d_ref:
@@ -131,7 +162,8 @@ class FusedPrioritizer:
                 self.lsa_ready = True
 
     def _lsa_lse(self, padded, bp, tseg_cpu, tseg):
-        """Per-class whiten GEMMs + one grouped KDE launch."""
+        """Per-class whiten GEMMs + one grouped KDE launch (+ cross-rank
+        logsumexp merge of the train-shard partials in shard mode)."""
         white = torch.zeros(bp, self.lsa_d, device=padded.device)
         for c in range(self.num_classes):
             lo, hi = int(tseg_cpu[c]), int(tseg_cpu[c + 1])
@@ -142,10 +174,37 @@ class FusedPrioritizer:
             if keep is not None:
                 seg = seg.index_select(1, keep)
             white[lo:hi] = seg @ linv_t
-        return self.ext.grouped_kde(
+        lse = self.ext.grouped_kde(
             white.contiguous(), self.lsa_wtrainS, tseg, self.lsa_nseg,
             self.lsa_wnormS, self.lsa_jb_max,
         )
+        if self.shard_train:
+            import torch.distributed as tdist
+
+            world = tdist.get_world_size()
+            parts = [torch.empty_like(lse) for _ in range(world)]
+            tdist.all_gather(parts, lse.contiguous())
+            lse = torch.logsumexp(torch.stack(parts), dim=0)
+        return lse
+
+    @staticmethod
+    def _merge_rowmin(dist, idx):
+        """Strict-less rank-ordered merge of (min, global argmin) partials —
+        lowest rank holds the lowest global rows per class, so ties keep the
+        single-device lowest-index rule."""
+        import torch.distributed as tdist
+
+        world = tdist.get_world_size()
+        dg = [torch.empty_like(dist) for _ in range(world)]
+        ig = [torch.empty_like(idx) for _ in range(world)]
+        tdist.all_gather(dg, dist.contiguous())
+        tdist.all_gather(ig, idx.contiguous())
+        best_d, best_i = dg[0], ig[0]
+        for r in range(1, world):
+            take = dg[r] < best_d
+            best_d = torch.where(take, dg[r], best_d)
+            best_i = torch.where(take, ig[r], best_i)
+        return best_d, best_i
 
     def _segment(self, ats: torch.Tensor, pred: torch.Tensor):
         """Class-sort + pad to 128-row segments."""
@@ -196,6 +255,11 @@ class FusedPrioritizer:
             padded.contiguous(), self.trainS, tseg, self.nseg, self.bnormS,
             self.jb_max,
         )
+        if self.shard_train:
+            idx = torch.where(
+                idx >= 0, self.local2global[idx.clamp_min(0)], idx
+            )
+            dist, idx = self._merge_rowmin(dist, idx)
         dsa_sorted = torch.where(
             idx >= 0,
             dist / self.btableS[idx.clamp_min(0)],

@@ -36,8 +36,22 @@ class SurpriseHandler:
         ),
     }
 
-    def __init__(self, model, sa_layers: List[int], training_dataset, device=None, predict_batch: int = 512):
+    def __init__(
+        self,
+        model,
+        sa_layers: List[int],
+        training_dataset,
+        device=None,
+        predict_batch: int = 512,
+        dist_shard: bool = False,
+    ):
+        from ..parallel.dist import get_world_size
+
         self.sa_layers = list(sa_layers)
+        # dist_shard: shard AT-extraction and SA scoring over the test-input
+        # axis across ranks (DP per SURVEY §2.4); train ATs are extracted
+        # sharded then all-gathered so every rank fits identical SAs.
+        self.dist_shard = bool(dist_shard) and get_world_size() > 1
         self.base_model = BaseModel(
             model,
             self.sa_layers,
@@ -47,7 +61,19 @@ class SurpriseHandler:
         )
         self.train_at_timer = DeviceTimer()
         with self.train_at_timer:
-            self.train_ats, self.train_pred = self._acti_and_pred(training_dataset)
+            if self.dist_shard:
+                from ..parallel.dist import allgather_rows, shard_slice
+
+                n = training_dataset.shape[0]
+                ats_l, pred_l = self._acti_and_pred(
+                    training_dataset[shard_slice(n)]
+                )
+                self.train_ats = allgather_rows(ats_l, n)
+                self.train_pred = allgather_rows(pred_l, n)
+            else:
+                self.train_ats, self.train_pred = self._acti_and_pred(
+                    training_dataset
+                )
 
     def _acti_and_pred(self, dataset):
         """ATs and argmax predictions in one fused forward pass (K15)."""
@@ -66,8 +92,17 @@ class SurpriseHandler:
         for ds_name, dataset in datasets.items():
             t = DeviceTimer()
             with t:
-                test_ats, test_pred = self._acti_and_pred(dataset)
-            test_apt[ds_name] = (test_ats, test_pred, t.get())
+                if self.dist_shard:
+                    from ..parallel.dist import shard_slice
+
+                    n = dataset.shape[0]
+                    test_ats, test_pred = self._acti_and_pred(
+                        dataset[shard_slice(n)]
+                    )
+                else:
+                    n = None
+                    test_ats, test_pred = self._acti_and_pred(dataset)
+            test_apt[ds_name] = (test_ats, test_pred, t.get(), n)
 
         for sa_name, sa_func in self.TESTED_SA.items():
             res[sa_name] = {}
@@ -79,11 +114,16 @@ class SurpriseHandler:
                     sa.badge_size = dsa_badge_size
             setup_time = self.train_at_timer.get() + setup_timer.get()
 
-            for ds_name, (test_ats, test_pred, pred_time) in test_apt.items():
+            for ds_name, (test_ats, test_pred, pred_time, n) in test_apt.items():
                 sa_timer = DeviceTimer()
                 with sa_timer:
                     logger.info("Calculating %s for %s", sa_name, ds_name)
                     sa_vals = sa(test_ats, test_pred)
+                    if self.dist_shard:
+                        # publish the per-input score shard (tiny all-gather)
+                        from ..parallel.dist import allgather_rows
+
+                        sa_vals = allgather_rows(sa_vals.contiguous(), n)
                 res[sa_name][ds_name] = (sa_vals, [setup_time, pred_time, sa_timer.get()])
 
         for sa_name in self.TESTED_SA.keys():

@@ -70,9 +70,17 @@ def shard_slice(n: int, rank: Optional[int] = None, world: Optional[int] = None)
 
 def allgather_rows(local: torch.Tensor, n_total: int) -> torch.Tensor:
     """All-gather row shards produced by :func:`shard_slice` into the full
-    [n_total, ...] tensor (identical on every rank)."""
+    [n_total, ...] tensor (identical on every rank).
+
+    The nccl (RCCL) backend moves CPU shards through the device; the result
+    comes back on the input's device.
+    """
     if not is_initialized():
         return local
+    home = None
+    if dist.get_backend() == "nccl" and not local.is_cuda:
+        home = local.device
+        local = local.cuda()
     world = get_world_size()
     pad = (n_total + world - 1) // world  # equal-size buffers (RCCL-safe)
     buf = torch.zeros((pad,) + tuple(local.shape[1:]), dtype=local.dtype,
@@ -84,7 +92,8 @@ def allgather_rows(local: torch.Tensor, n_total: int) -> torch.Tensor:
     for r in range(world):
         s = shard_slice(n_total, r, world)
         parts.append(shards[r][: s.stop - s.start])
-    return torch.cat(parts, dim=0)
+    out = torch.cat(parts, dim=0)
+    return out.to(home) if home is not None else out
 
 
 def allreduce_max_scalar(value: float, device) -> float:

@@ -26,15 +26,26 @@ def shard_rows(full: torch.Tensor) -> Tuple[torch.Tensor, int]:
 
 
 def sharded_rowmin_l2(
-    test: torch.Tensor, train_shard: torch.Tensor, shard_offset: int
+    test: torch.Tensor,
+    train_shard: torch.Tensor,
+    shard_offset: int,
+    train_shard_norm: torch.Tensor = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Global (min L2 distance, global argmin) with the train rows sharded.
 
     Partials merge in rank order with strict-less comparison, preserving the
-    lowest-global-index tie rule of the single-device path.
+    lowest-global-index tie rule of the single-device path. A rank whose
+    shard is empty (class smaller than the world size) contributes +inf
+    partials and never wins the merge.
     """
-    d, i = ops.rowmin_l2(test, train_shard)
-    i = i + shard_offset
+    if train_shard.shape[0] == 0:
+        d = torch.full(
+            (test.shape[0],), float("inf"), dtype=test.dtype, device=test.device
+        )
+        i = torch.zeros(test.shape[0], dtype=torch.int64, device=test.device)
+    else:
+        d, i = ops.rowmin_l2(test, train_shard, train_shard_norm)
+        i = i + shard_offset
     if not is_initialized():
         return d, i
     world = get_world_size()
@@ -58,8 +69,14 @@ def sharded_kde_logsumexp(
 
     Per-rank partial LSEs merge with a rank-ordered streaming logsumexp —
     the partial-reduction pattern this workload has in place of
-    ring-attention."""
-    part = ops.kde_logsumexp(test_w, train_shard_w)
+    ring-attention. Empty shards contribute -inf partials (the identity)."""
+    if train_shard_w.shape[0] == 0:
+        part = torch.full(
+            (test_w.shape[0],), float("-inf"),
+            dtype=test_w.dtype, device=test_w.device,
+        )
+    else:
+        part = ops.kde_logsumexp(test_w, train_shard_w)
     if not is_initialized():
         return part
     world = get_world_size()
@@ -69,12 +86,50 @@ def sharded_kde_logsumexp(
     return torch.logsumexp(stacked, dim=0)
 
 
+def allreduce_bitmap_or(words: torch.Tensor) -> torch.Tensor:
+    """Bitwise-OR all-reduce of packed int64 coverage-bitmap words, in place.
+
+    The coverage-sharding collective (BASELINE config 4): each rank holds the
+    profile rows of its test-input shard in a full-size [N, W] matrix (zeros
+    outside its shard) and the OR union reassembles the full profile matrix
+    identically on every rank. RCCL, like NCCL, exposes no BOR reduction op,
+    so on the nccl backend the OR is an all-gather over xGMI followed by a
+    deterministic local fold — these messages are KB-to-MB, latency-bound on
+    7x153 GB/s xGMI either way. gloo carries BOR natively.
+    """
+    if not is_initialized():
+        return words
+    if dist.get_backend() == "gloo":
+        dist.all_reduce(words, op=dist.ReduceOp.BOR)
+        return words
+    world = get_world_size()
+    parts = [torch.empty_like(words) for _ in range(world)]
+    dist.all_gather(parts, words.contiguous())
+    acc = parts[0]
+    for r in range(1, world):
+        acc = torch.bitwise_or(acc, parts[r])
+    words.copy_(acc)
+    return words
+
+
 def allreduce_minmax(mins: torch.Tensor, maxs: torch.Tensor):
     """Cross-rank elementwise min/max of aggregate statistics (K18)."""
     if is_initialized():
         dist.all_reduce(mins, op=dist.ReduceOp.MIN)
         dist.all_reduce(maxs, op=dist.ReduceOp.MAX)
     return mins, maxs
+
+
+def allreduce_min(t: torch.Tensor) -> torch.Tensor:
+    if is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    return t
+
+
+def allreduce_max(t: torch.Tensor) -> torch.Tensor:
+    if is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return t
 
 
 def allreduce_welford(count: float, mean: torch.Tensor, m2: torch.Tensor):
