@@ -440,9 +440,24 @@ def main():
     _, dsa_last, lsa_last, _ = last
     apfd_dsa = _apfd(dsa_last)
     apfd_lsa = _apfd(lsa_last) if lsa_last is not None else float("nan")
+    # quality pin: on the NOMINAL half of the batch a
+    # trained model's deep-gini ordering must beat random — rules out a
+    # sign/negation bug producing a plausible-looking throughput number
+    half = args.batch // 2
+    mis_nom = mis[:half]
+    apfd_gini_nom = float("nan")
+    if mis_nom.sum() >= 10:
+        apfd_gini_nom = apfd_from_order(
+            mis_nom,
+            np.argsort(-gini[:half].float().cpu().numpy(), kind="stable"),
+        )
+        assert apfd_gini_nom > 0.5, (
+            f"deep-gini nominal APFD {apfd_gini_nom:.3f} <= 0.5: softmax "
+            f"ordering is not ranking real faults"
+        )
     log(rank, f"last-batch accuracy={1.0 - mis.mean():.3f} "
-              f"apfd_gini={apfd:.3f} apfd_dsa={apfd_dsa:.3f} "
-              f"apfd_lsa={apfd_lsa:.3f}")
+              f"apfd_gini={apfd:.3f} apfd_gini_nominal={apfd_gini_nom:.3f} "
+              f"apfd_dsa={apfd_dsa:.3f} apfd_lsa={apfd_lsa:.3f}")
 
     if rank == 0:
         print(
@@ -471,6 +486,9 @@ def main():
                         "at_width": 4096,
                         "scorers": "gini+softmax-family+dsa+pc-lsa",
                         "apfd_gini_lastbatch": None if np.isnan(apfd) else apfd,
+                        "apfd_gini_nominal_lastbatch": (
+                            None if np.isnan(apfd_gini_nom) else apfd_gini_nom
+                        ),
                         "apfd_dsa_lastbatch": None if np.isnan(apfd_dsa) else apfd_dsa,
                         "apfd_pclsa_lastbatch": None if np.isnan(apfd_lsa) else apfd_lsa,
                     },

@@ -94,6 +94,10 @@ def silhouette_score(x: torch.Tensor, labels: torch.Tensor, chunk: int = 2048) -
             own_counts - 1
         ).clamp_min(1.0)
         mean_to = sums / counts.clamp_min(1.0)
+        # empty clusters (kmeans_fit can retain a center nobody claims) must
+        # not offer b=0 to every sample — sklearn raises on such labelings;
+        # masking them to +inf drops them from the min
+        mean_to[:, counts == 0] = float("inf")
         mean_to[torch.arange(xc.shape[0], device=x.device), own] = float("inf")
         b = mean_to.min(dim=1).values
         s = (b - a) / torch.maximum(a, b)

@@ -16,10 +16,10 @@ same behavioural contract:
   the extension; pure-python fallback for small dictionaries).
 
 Deviation (no network egress in this environment): the reference downloads
-a WordNet thesaurus for SYNONYM (text_corruptor.py:412-446). Here synonyms
-are an injectable dict; without one, SYNONYM falls back to AUTOCORRECT
-(nearest dictionary word), which the reference also does whenever a word
-has no synonyms.
+a WordNet thesaurus for SYNONYM (text_corruptor.py:412-446). Here a compact
+curated synonym table ships in-tree (synonyms_data.py) and is used by
+default; a custom dict stays injectable. Words without synonyms fall back
+to AUTOCORRECT (nearest dictionary word), which the reference also does.
 """
 
 import collections
@@ -106,7 +106,11 @@ class TextCorruptor:
         self.word_index = {w: i for i, w in enumerate(self.common_words)}
         self.start_bags = self._word_start_bags()
         self.distances = self._calculate_distances()
-        self.synonyms = synonyms or {}
+        if synonyms is None:
+            from .synonyms_data import SYNONYMS
+
+            synonyms = SYNONYMS
+        self.synonyms = synonyms
 
     # -- dictionary construction ----------------------------------------
 

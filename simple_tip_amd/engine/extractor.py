@@ -1,0 +1,224 @@
+"""Graph-captured AT extraction for the engine (K15 fast path).
+
+Round-1 profiling (profiles/r01_bench_kernels.md) showed the eager
+MIOpen-backed `forward_taps` is LAUNCH-bound for these small models
+(~361 us/launch, BN inference kernels dominating on ResNet-20). This module
+promotes the bench-only optimisation into the engine:
+
+- BatchNorm folds into the preceding convs (exact algebra, models/fuse.py);
+- image models run bf16 + channels_last with the stem padded to >= 4
+  channels (MIOpen's NHWC bf16 igemm requirement);
+- ResNet-20 routes to the hand-written fused block kernels
+  (ops/hip/resnet_fused.hip) when the extension is available;
+- the fixed-shape forward is captured ONCE into a hipGraph and replayed per
+  batch — per-step launch overhead becomes one graph launch.
+
+Batches that do not match the captured shape (the dataset remainder) take
+the eager folded path. Taps are returned as fp32 copies (replay reuses the
+static buffers), softmax probabilities as an fp32 clone.
+
+Reference hot path being replaced: handler_model.py:175-206 (the
+"transparent model" predict).
+"""
+
+import logging
+import os
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+import torch.nn as nn
+
+logger = logging.getLogger(__name__)
+
+
+def _first_conv(model: nn.Module) -> Optional[nn.Conv2d]:
+    for m in model.modules():
+        if isinstance(m, nn.Conv2d):
+            return m
+    return None
+
+
+class GraphedExtractor:
+    """hipGraph-replayed, BN-folded AT extraction for one (model, taps)."""
+
+    def __init__(
+        self,
+        model,
+        activation_layers: Optional[Sequence[int]],
+        device: torch.device,
+        batch: int,
+        use_graph: bool = True,
+    ):
+        from ..models.fuse import fold_bn_inference, pad_stem_channels
+
+        self.device = device
+        self.batch = int(batch)
+        self.layers = list(activation_layers or [])
+        m = fold_bn_inference(model).to(device)
+        conv = _first_conv(m)
+        self.image_mode = conv is not None
+        self.in_ch = 0
+        self.fused = None
+        if self.image_mode:
+            if conv.in_channels < 4:
+                pad_stem_channels(m, 4)
+            self.in_ch = _first_conv(m).in_channels
+            self.model = m.to(torch.bfloat16).to(
+                memory_format=torch.channels_last
+            )
+        else:
+            # token/sequence models: keep fp32 weights (embeddings +
+            # attention are tiny; the win here is the graph replay)
+            self.model = m
+        self._try_fused_resnet(model)
+        self.graph = None
+        if use_graph and device.type == "cuda":
+            try:
+                self._capture()
+            except Exception as e:  # pragma: no cover - capture is optional
+                logger.warning("hipGraph capture failed (%r); eager path", e)
+                self.graph = None
+
+    def _try_fused_resnet(self, orig_model):
+        """Route ResNet-20 through the hand-written fused block kernels."""
+        from ..models.cnn import ResNet20
+
+        if (
+            not isinstance(orig_model, ResNet20)
+            or self.device.type != "cuda"
+            or os.environ.get("TIP_NO_FUSED_RESNET") == "1"
+            or self.layers != list(ResNet20.sa_layers)
+        ):
+            return
+        try:
+            from ..models.fuse import fold_bn_inference
+            from ..models.resnet_fused import FusedResNet20
+
+            self.fused = FusedResNet20(
+                fold_bn_inference(orig_model).to(self.device), self.device
+            )
+        except Exception as e:  # pragma: no cover
+            logger.warning("fused ResNet unavailable (%r)", e)
+            self.fused = None
+
+    # -- shape handling --------------------------------------------------
+
+    def _prep(self, xb: torch.Tensor) -> torch.Tensor:
+        xb = xb.to(self.device, non_blocking=True)
+        if not self.image_mode:
+            return xb
+        xb = xb.to(torch.bfloat16)
+        if xb.shape[1] < self.in_ch:
+            pad = torch.zeros(
+                xb.shape[0], self.in_ch - xb.shape[1], *xb.shape[2:],
+                dtype=xb.dtype, device=xb.device,
+            )
+            xb = torch.cat([xb, pad], dim=1)
+        return xb.to(memory_format=torch.channels_last)
+
+    # -- graph capture ----------------------------------------------------
+
+    @torch.no_grad()
+    def _capture(self):
+        if self.fused is not None:
+            self._capture_fused()
+            return
+        probe = torch.zeros(
+            self.batch,
+            self.in_ch if self.image_mode else 1,
+            *self._spatial_shape(),
+            device=self.device,
+            dtype=torch.bfloat16 if self.image_mode else torch.long,
+        )
+        if not self.image_mode:
+            probe = probe.reshape(self.batch, -1)
+        else:
+            probe = probe.to(memory_format=torch.channels_last)
+        self.static_x = probe
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):  # warm MIOpen finds before capture
+                self.model.forward_taps(self.static_x, self.layers)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            taps, logits = self.model.forward_taps(self.static_x, self.layers)
+            self.static_taps = list(taps)
+            self.static_probs = torch.softmax(logits.float(), dim=1)
+        self.graph = g
+
+    @torch.no_grad()
+    def _capture_fused(self):
+        """Capture the fused-ResNet forward (NHWC fp32 input)."""
+        self.fstatic_x = torch.zeros(self.batch, 32, 32, 3, device=self.device)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self.fused.forward_nhwc(self.fstatic_x)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            ats, logits = self.fused.forward_nhwc(self.fstatic_x)
+            self.static_taps = [ats]
+            self.static_probs = torch.softmax(logits.float(), dim=1)
+        self.graph = g
+
+    def _spatial_shape(self):
+        # the capture probe only needs A valid fixed shape; real batches of
+        # other spatial sizes fall back to eager
+        m = self.model
+        shape = getattr(m, "input_shape", None)
+        if shape is None:
+            return (32, 32) if self.image_mode else (100,)
+        return tuple(shape[1:]) if self.image_mode else tuple(shape)
+
+    # -- extraction --------------------------------------------------------
+
+    @torch.no_grad()
+    def __call__(self, xb) -> Tuple[List[torch.Tensor], torch.Tensor]:
+        """(fp32 tap copies, fp32 softmax) for one batch."""
+        if not isinstance(xb, torch.Tensor):
+            import numpy as np
+
+            xb = torch.from_numpy(np.ascontiguousarray(xb))
+        if xb.dtype == torch.float64:
+            xb = xb.float()
+        if self.fused is not None and self.graph is not None:
+            if xb.shape[0] == self.batch and xb.dim() == 4:
+                # fused path consumes NHWC fp32
+                nhwc = (
+                    xb if xb.shape[-1] == 3 else xb.permute(0, 2, 3, 1)
+                ).contiguous()
+                self.fstatic_x.copy_(nhwc.to(self.device, torch.float32))
+                self.graph.replay()
+                return (
+                    [t.float() if t.dtype != torch.float32 else t.clone()
+                     for t in self.static_taps],
+                    self.static_probs.clone(),
+                )
+        elif self.fused is not None:
+            nhwc = (
+                xb if xb.dim() == 4 and xb.shape[-1] == 3
+                else xb.permute(0, 2, 3, 1)
+            ).contiguous()
+            ats, logits = self.fused.forward_nhwc(
+                nhwc.to(self.device, torch.float32)
+            )
+            return [ats], torch.softmax(logits.float(), dim=1)
+        if self.graph is not None and xb.shape[0] == self.batch:
+            prepped = self._prep(xb)
+            if prepped.shape == self.static_x.shape:
+                self.static_x.copy_(prepped)
+                self.graph.replay()
+                return (
+                    [t.float() if t.dtype != torch.float32 else t.clone()
+                     for t in self.static_taps],
+                    self.static_probs.clone(),
+                )
+        taps, logits = self.model.forward_taps(self._prep(xb), self.layers)
+        return (
+            [t.float() for t in taps],
+            torch.softmax(logits.float(), dim=1),
+        )

@@ -49,6 +49,7 @@ class BaseModel:
         self.include_last_layer = include_last_layer
         self.device = device or next(model.parameters()).device
         self.predict_batch = predict_batch
+        self._graphed = None  # lazy GraphedExtractor (GPU fast path, K15)
         self.model.eval()
 
     def _to_device(self, x) -> torch.Tensor:
@@ -76,9 +77,36 @@ class BaseModel:
 
     @torch.no_grad()
     def walk_activations(self, batches) -> Generator[List[torch.Tensor], None, None]:
-        """Stream batches -> per-batch list of tapped activation tensors."""
+        """Stream batches -> per-batch list of tapped activation tensors.
+
+        On GPU this routes through the BN-folded, hipGraph-replayed
+        :class:`~simple_tip_amd.engine.extractor.GraphedExtractor` (bf16
+        channels_last; fused ResNet-20 block kernels where applicable) —
+        the eager MIOpen path was launch-bound.
+        ``TIP_NO_GRAPH_EXTRACTOR=1`` restores the eager path for debugging.
+        """
+        import os
+
         if self.activation_layers is None:
             raise ValueError("No activation layers specified")
+        if (
+            self.device.type == "cuda"
+            and os.environ.get("TIP_NO_GRAPH_EXTRACTOR") != "1"
+        ):
+            if self._graphed is None:
+                from .extractor import GraphedExtractor
+
+                self._graphed = GraphedExtractor(
+                    self.model, self.activation_layers, self.device,
+                    self.predict_batch,
+                )
+            for batch in batches:
+                taps, probs = self._graphed(batch)
+                out = list(taps)
+                if self.include_last_layer:
+                    out.append(probs)
+                yield out
+            return
         for batch in batches:
             xb = self._to_device(batch)
             taps, logits = self.model.forward_taps(xb, self.activation_layers)

@@ -6,8 +6,10 @@ Pools per-(setting, run) samples across settings, then for every approach
 pair computes a two-sided Wilcoxon signed-rank p (Bonferroni-corrected by
 C(39, 2)) and a paired Vargha-Delaney A12 folded to [0, 1]
 (reference correlation_plot.py:22-45). Emits
-results/{exp}_correlation_p.csv / _eff.csv; the heatmap figure is produced
-only when matplotlib is importable (not shipped in this image)."""
+results/{exp}_correlation_p.csv / _eff.csv plus the paper's dual-triangle
+heatmap figure (p-values upper triangle, effect sizes lower triangle —
+reference correlation_plot.py:116-183, Figs. 3-4) as
+results/{exp}_correlation.png."""
 
 import logging
 import os
@@ -81,22 +83,56 @@ def _write(exp: str, approaches, p, e):
         logger.info("matplotlib not available; skipping heatmap figure")
 
 
-def _plot_heatmap(exp, approaches, p, e):  # pragma: no cover - optional dep
+def _plot_heatmap(exp, approaches, p, e):
+    """Dual-triangle heatmap: Bonferroni p upper / folded A12 effect lower
+    (reference correlation_plot.py:116-183). Pure matplotlib; the reference
+    uses two masked seaborn heatmaps on one axes."""
     import matplotlib
 
     matplotlib.use("Agg")
     import matplotlib.pyplot as plt
 
-    fig, ax = plt.subplots(figsize=(8, 7))
-    mat = np.where(np.isnan(p), np.nan, p)
-    im = ax.imshow(mat, cmap="viridis")
-    ax.set_xticks(range(len(approaches)))
+    n = len(approaches)
+    # upper triangle: p (i<j as computed); lower: effect mirrored to (j,i)
+    p_mat = np.full((n, n), np.nan)
+    e_mat = np.full((n, n), np.nan)
+    iu = np.triu_indices(n, 1)
+    p_mat[iu] = p[iu]
+    e_mat[(iu[1], iu[0])] = e[iu]
+
+    fig, ax = plt.subplots(figsize=(1.1 * n + 2, 1.0 * n + 1.5))
+    im_p = ax.imshow(
+        np.ma.masked_invalid(p_mat), cmap="Reds_r", vmin=0.0, vmax=1.0
+    )
+    im_e = ax.imshow(
+        np.ma.masked_invalid(e_mat), cmap="Blues", vmin=0.0, vmax=1.0
+    )
+    ax.set_xticks(range(n))
     ax.set_xticklabels(approaches, rotation=90)
-    ax.set_yticks(range(len(approaches)))
+    ax.set_yticks(range(n))
     ax.set_yticklabels(approaches)
-    fig.colorbar(im)
+    for i in range(n):  # grey diagonal
+        ax.add_patch(
+            plt.Rectangle((i - 0.5, i - 0.5), 1, 1, color="0.85", lw=0)
+        )
+    for i in range(n):
+        for j in range(n):
+            v = p_mat[i, j] if j > i else e_mat[i, j]
+            if np.isfinite(v):
+                ax.text(
+                    j, i, f"{v:.2f}", ha="center", va="center", fontsize=8,
+                    color="black",
+                )
+    cb_p = fig.colorbar(im_p, ax=ax, fraction=0.045, pad=0.02)
+    cb_p.set_label("Wilcoxon p (Bonferroni), upper")
+    cb_e = fig.colorbar(im_e, ax=ax, fraction=0.045, pad=0.08)
+    cb_e.set_label("Vargha-Delaney effect (folded), lower")
+    ax.set_title(f"{exp}: pairwise significance and effect size")
     fig.tight_layout()
-    fig.savefig(os.path.join(config.OUTPUT_FOLDER, "results", f"{exp}_correlation.png"))
+    fig.savefig(
+        os.path.join(config.OUTPUT_FOLDER, "results", f"{exp}_correlation.png"),
+        dpi=120,
+    )
     plt.close(fig)
 
 

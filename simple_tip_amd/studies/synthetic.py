@@ -44,12 +44,28 @@ def synthetic_images(
     shape: Tuple[int, ...],
     num_classes: int,
     noise: float = 0.8,
+    ambiguous: float = 0.08,
 ) -> Tuple[np.ndarray, np.ndarray]:
-    """(x, y): class template + Gaussian noise, scaled to roughly [0, 1]."""
+    """(x, y): class template + Gaussian noise, scaled to roughly [0, 1].
+
+    An ``ambiguous`` fraction of samples blends the labeled class's template
+    ~50/50 with another class's, making them irreducibly hard: a trained
+    model misclassifies about half of them (a few % overall), so nominal
+    test sets contain real faults and nominal APFD is well-defined — the
+    reference's real datasets have this property intrinsically
+    (reference eval_apfd_table.py:111-131 expects a
+    populated nominal column)."""
     rng = _rng(study, split)
     y = rng.randint(0, num_classes, size=n)
     templates = image_class_templates(study, shape, num_classes)
-    x = templates[y] + noise * rng.randn(n, *shape).astype(np.float32)
+    base = templates[y].copy()
+    if ambiguous > 0 and num_classes > 1:
+        m = rng.rand(n) < ambiguous
+        alt = (y + rng.randint(1, num_classes, size=n)) % num_classes
+        w = (0.45 + 0.10 * rng.rand(n)).astype(np.float32)
+        wm = w[m].reshape((-1,) + (1,) * len(shape))
+        base[m] = (1.0 - wm) * base[m] + wm * templates[alt[m]]
+    x = base + noise * rng.randn(n, *shape).astype(np.float32)
     x = (x - x.min()) / (x.max() - x.min() + 1e-8)
     return x.astype(np.float32), y.astype(np.int64)
 
@@ -99,11 +115,22 @@ def synthetic_tokens(
     base = 1.0 / (np.arange(vocab_size) + 10.0)
     class_boost = _rng(study, "token-classes").rand(num_classes, vocab_size) ** 4
     x = np.empty((n, seq_len), dtype=np.int64)
+    class_probs = np.empty((num_classes, vocab_size))
     for c in range(num_classes):
         probs = base * (1.0 + 8.0 * class_boost[c])
-        probs /= probs.sum()
+        class_probs[c] = probs / probs.sum()
         sel = y == c
-        x[sel] = rng.choice(vocab_size, size=(int(sel.sum()), seq_len), p=probs)
+        x[sel] = rng.choice(vocab_size, size=(int(sel.sum()), seq_len), p=class_probs[c])
+    # ambiguous fraction: tokens drawn from a ~50/50 two-class mixture, so
+    # trained models have an irreducible nominal error (see synthetic_images)
+    ambiguous = 0.08
+    if num_classes > 1:
+        m = np.nonzero(rng.rand(n) < ambiguous)[0]
+        alt = (y[m] + rng.randint(1, num_classes, size=m.shape[0])) % num_classes
+        w = 0.45 + 0.10 * rng.rand(m.shape[0])
+        for j, i in enumerate(m):
+            probs = (1.0 - w[j]) * class_probs[y[i]] + w[j] * class_probs[alt[j]]
+            x[i] = rng.choice(vocab_size, size=seq_len, p=probs)
     return x, y.astype(np.int64)
 
 

@@ -88,3 +88,21 @@ def test_pack_conv_fragment_math():
                     assert got == pytest.approx(
                         float(torch.tensor(want).to(torch.bfloat16)), abs=1e-6
                     ), (ct, ks, lane, e)
+
+
+def test_silhouette_ignores_empty_clusters():
+    """Empty clusters (kmeans can retain unclaimed centers) must not feed
+    b=0 into the min: score must equal sklearn's on the non-empty clusters
+    (ADVICE r01)."""
+    from sklearn.metrics import silhouette_score as sk_sil
+
+    from simple_tip_amd.core.kmeans import silhouette_score
+
+    torch.manual_seed(3)
+    x = torch.randn(40, 5) + torch.cat(
+        [torch.zeros(20, 5), torch.full((20, 5), 4.0)]
+    )
+    labels = torch.cat([torch.zeros(20), torch.full((20,), 2)]).long()
+    got = silhouette_score(x, labels)  # cluster id 1 is empty
+    want = float(sk_sil(x.numpy(), labels.numpy()))
+    assert abs(got - want) < 1e-5

@@ -111,3 +111,26 @@ def test_correlation_outputs(assets):
     assert os.path.exists(assets / "results" / "apfd_correlation_p.csv")
     assert 0 <= p[0, 1] <= 1
     assert 0 <= e[0, 1] <= 1
+
+
+def test_correlation_heatmap_png(tmp_path, monkeypatch):
+    """Fig 3/4 parity: the dual-triangle heatmap renders to PNG from
+    synthetic measurement dicts (reference
+    correlation_plot.py:116-183)."""
+    import numpy as np
+
+    from simple_tip_amd import config
+    from simple_tip_amd.results import correlation
+
+    monkeypatch.setattr(config, "OUTPUT_FOLDER", str(tmp_path))
+    approaches = ["dsa", "pc-lsa", "deep_gini", "softmax"]
+    rng = np.random.RandomState(0)
+    measurements = {
+        a: {f"s:{k}": float(rng.rand()) for k in range(12)} for a in approaches
+    }
+    p, e = correlation._pairwise(measurements, approaches)
+    assert np.isfinite(p[0, 1]) and np.isfinite(e[0, 1])
+    correlation._write("testexp", approaches, p, e)
+    png = tmp_path / "results" / "testexp_correlation.png"
+    assert png.exists() and png.stat().st_size > 5000
+    assert (tmp_path / "results" / "testexp_correlation_p.csv").exists()

@@ -100,3 +100,23 @@ def test_weights_all_types_reachable(corruptor):
     for orig, new in zip(text.split(), out.split()):
         if new != orig:
             assert len(new) == len(orig)
+
+
+def test_synonym_table_ships_in_tree():
+    """SYNONYM (weight 0.35) must exercise its own path via the bundled
+    thesaurus instead of silently degrading to AUTOCORRECT (VERDICT r01
+    item 8; reference downloads WordNet, text_corruptor.py:412-446)."""
+    from simple_tip_amd.core.synonyms_data import SYNONYMS
+    from simple_tip_amd.core.text_corruptor import TextCorruptor
+
+    assert len(SYNONYMS) >= 250
+    assert all(s and all(isinstance(w, str) for w in s) for s in SYNONYMS.values())
+
+    base = ["a wonderful story about a terrible movie"] * 5
+    tc = TextCorruptor(base_dataset=base, dict_size=100)
+    out = tc._corrupt_synonym("wonderful", seed=1)
+    assert out in SYNONYMS["wonderful"]
+    # determinism
+    assert out == tc._corrupt_synonym("wonderful", seed=1)
+    # unknown words still fall back to autocorrect semantics
+    assert tc._corrupt_synonym("qqqqq", seed=1) == tc._corrupt_autocorrect("qqqqq", seed=1)
