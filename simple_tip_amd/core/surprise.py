@@ -490,8 +490,13 @@ class MLSA(SA):
             return
         import torch.distributed as dist
 
+        from ..parallel.dist import _staging_device
+
         for t in (self.means, self.prec_chol, self.log_weights):
-            dist.broadcast(t, src=0)
+            staged, home = _staging_device(t)
+            dist.broadcast(staged, src=0)
+            if home is not None:
+                t.copy_(staged.to(home))
 
     def _fit_device(self, x: torch.Tensor, k: int, iters: int = 60, tol: float = 1e-3):
         """Full-covariance EM in fp64 on the device (kmeans init)."""

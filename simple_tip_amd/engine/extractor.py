@@ -6,12 +6,19 @@ MIOpen-backed `forward_taps` is LAUNCH-bound for these small models
 promotes the bench-only optimisation into the engine:
 
 - BatchNorm folds into the preceding convs (exact algebra, models/fuse.py);
-- image models run bf16 + channels_last with the stem padded to >= 4
-  channels (MIOpen's NHWC bf16 igemm requirement);
 - ResNet-20 routes to the hand-written fused block kernels
   (ops/hip/resnet_fused.hip) when the extension is available;
 - the fixed-shape forward is captured ONCE into a hipGraph and replayed per
   batch — per-step launch overhead becomes one graph launch.
+
+Engine extraction computes in fp32 by default (the reference's dtype):
+coverage boundary metrics (NBC/SNAC scaler 0) test equality against train
+extremes, and bf16 rounding makes unrelated test activations collide with
+those boundaries (measured: 100% of NBC_0 scores shifted). The launch-bound
+cost the graph removes is dtype-independent. ``TIP_EXTRACTOR_BF16=1`` opts
+image models into bf16 + channels_last (stem padded to >= 4 channels for
+MIOpen's NHWC bf16 igemm) where throughput matters more than boundary
+semantics — the bench uses its own bf16 pipeline either way.
 
 Batches that do not match the captured shape (the dataset remainder) take
 the eager folded path. Taps are returned as fp32 copies (replay reuses the
@@ -57,9 +64,12 @@ class GraphedExtractor:
         m = fold_bn_inference(model).to(device)
         conv = _first_conv(m)
         self.image_mode = conv is not None
+        self.bf16 = (
+            self.image_mode and os.environ.get("TIP_EXTRACTOR_BF16") == "1"
+        )
         self.in_ch = 0
         self.fused = None
-        if self.image_mode:
+        if self.bf16:
             if conv.in_channels < 4:
                 pad_stem_channels(m, 4)
             self.in_ch = _first_conv(m).in_channels
@@ -67,8 +77,7 @@ class GraphedExtractor:
                 memory_format=torch.channels_last
             )
         else:
-            # token/sequence models: keep fp32 weights (embeddings +
-            # attention are tiny; the win here is the graph replay)
+            self.in_ch = conv.in_channels if conv is not None else 0
             self.model = m
         self._try_fused_resnet(model)
         self.graph = None
@@ -105,7 +114,7 @@ class GraphedExtractor:
 
     def _prep(self, xb: torch.Tensor) -> torch.Tensor:
         xb = xb.to(self.device, non_blocking=True)
-        if not self.image_mode:
+        if not self.bf16:
             return xb
         xb = xb.to(torch.bfloat16)
         if xb.shape[1] < self.in_ch:
@@ -123,17 +132,19 @@ class GraphedExtractor:
         if self.fused is not None:
             self._capture_fused()
             return
-        probe = torch.zeros(
-            self.batch,
-            self.in_ch if self.image_mode else 1,
-            *self._spatial_shape(),
-            device=self.device,
-            dtype=torch.bfloat16 if self.image_mode else torch.long,
-        )
-        if not self.image_mode:
-            probe = probe.reshape(self.batch, -1)
+        if self.image_mode:
+            probe = torch.zeros(
+                self.batch, self.in_ch, *self._spatial_shape(),
+                device=self.device,
+                dtype=torch.bfloat16 if self.bf16 else torch.float32,
+            )
+            if self.bf16:
+                probe = probe.to(memory_format=torch.channels_last)
         else:
-            probe = probe.to(memory_format=torch.channels_last)
+            probe = torch.zeros(
+                self.batch, *self._spatial_shape(),
+                device=self.device, dtype=torch.long,
+            )
         self.static_x = probe
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())

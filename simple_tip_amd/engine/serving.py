@@ -179,12 +179,9 @@ class FusedPrioritizer:
             self.lsa_wnormS, self.lsa_jb_max,
         )
         if self.shard_train:
-            import torch.distributed as tdist
+            from ..parallel.dist import gather_tensors
 
-            world = tdist.get_world_size()
-            parts = [torch.empty_like(lse) for _ in range(world)]
-            tdist.all_gather(parts, lse.contiguous())
-            lse = torch.logsumexp(torch.stack(parts), dim=0)
+            lse = torch.logsumexp(torch.stack(gather_tensors(lse)), dim=0)
         return lse
 
     @staticmethod
@@ -192,13 +189,11 @@ class FusedPrioritizer:
         """Strict-less rank-ordered merge of (min, global argmin) partials —
         lowest rank holds the lowest global rows per class, so ties keep the
         single-device lowest-index rule."""
-        import torch.distributed as tdist
+        from ..parallel.dist import gather_tensors, get_world_size
 
-        world = tdist.get_world_size()
-        dg = [torch.empty_like(dist) for _ in range(world)]
-        ig = [torch.empty_like(idx) for _ in range(world)]
-        tdist.all_gather(dg, dist.contiguous())
-        tdist.all_gather(ig, idx.contiguous())
+        world = get_world_size()
+        dg = gather_tensors(dist)
+        ig = gather_tensors(idx)
         best_d, best_i = dg[0], ig[0]
         for r in range(1, world):
             take = dg[r] < best_d

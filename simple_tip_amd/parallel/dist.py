@@ -68,32 +68,46 @@ def shard_slice(n: int, rank: Optional[int] = None, world: Optional[int] = None)
     return slice(start, start + base + (1 if rank < rem else 0))
 
 
+def _staging_device(t: torch.Tensor):
+    """Where a tensor must live for this backend's collectives: nccl (RCCL)
+    needs device tensors; gloo implements all_gather only on host. Returns
+    (tensor staged for the backend, original device or None if unmoved)."""
+    backend = dist.get_backend()
+    if backend == "nccl" and not t.is_cuda:
+        return t.cuda(), t.device
+    if backend == "gloo" and t.is_cuda:
+        return t.cpu(), t.device
+    return t, None
+
+
+def gather_tensors(t: torch.Tensor) -> List[torch.Tensor]:
+    """all_gather equal-shaped tensors with backend/device staging; results
+    come back on the input's device, in rank order."""
+    world = get_world_size()
+    staged, home = _staging_device(t)
+    out = [torch.empty_like(staged) for _ in range(world)]
+    dist.all_gather(out, staged.contiguous())
+    if home is not None:
+        out = [o.to(home) for o in out]
+    return out
+
+
 def allgather_rows(local: torch.Tensor, n_total: int) -> torch.Tensor:
     """All-gather row shards produced by :func:`shard_slice` into the full
-    [n_total, ...] tensor (identical on every rank).
-
-    The nccl (RCCL) backend moves CPU shards through the device; the result
-    comes back on the input's device.
-    """
+    [n_total, ...] tensor (identical on every rank)."""
     if not is_initialized():
         return local
-    home = None
-    if dist.get_backend() == "nccl" and not local.is_cuda:
-        home = local.device
-        local = local.cuda()
     world = get_world_size()
     pad = (n_total + world - 1) // world  # equal-size buffers (RCCL-safe)
     buf = torch.zeros((pad,) + tuple(local.shape[1:]), dtype=local.dtype,
                       device=local.device)
     buf[: local.shape[0]] = local
-    shards = [torch.empty_like(buf) for _ in range(world)]
-    dist.all_gather(shards, buf)
+    shards = gather_tensors(buf)
     parts = []
     for r in range(world):
         s = shard_slice(n_total, r, world)
         parts.append(shards[r][: s.stop - s.start])
-    out = torch.cat(parts, dim=0)
-    return out.to(home) if home is not None else out
+    return torch.cat(parts, dim=0)
 
 
 def allreduce_max_scalar(value: float, device) -> float:

@@ -15,7 +15,14 @@ import torch
 import torch.distributed as dist
 
 from .. import ops
-from .dist import get_rank, get_world_size, is_initialized, shard_slice
+from .dist import (
+    _staging_device,
+    gather_tensors,
+    get_rank,
+    get_world_size,
+    is_initialized,
+    shard_slice,
+)
 
 
 def shard_rows(full: torch.Tensor) -> Tuple[torch.Tensor, int]:
@@ -49,10 +56,8 @@ def sharded_rowmin_l2(
     if not is_initialized():
         return d, i
     world = get_world_size()
-    dg = [torch.empty_like(d) for _ in range(world)]
-    ig = [torch.empty_like(i) for _ in range(world)]
-    dist.all_gather(dg, d.contiguous())
-    dist.all_gather(ig, i.contiguous())
+    dg = gather_tensors(d)
+    ig = gather_tensors(i)
     best_d, best_i = dg[0], ig[0]
     for r in range(1, world):
         # ranks hold ascending global offsets: strict less keeps lowest idx
@@ -79,10 +84,7 @@ def sharded_kde_logsumexp(
         part = ops.kde_logsumexp(test_w, train_shard_w)
     if not is_initialized():
         return part
-    world = get_world_size()
-    parts = [torch.empty_like(part) for _ in range(world)]
-    dist.all_gather(parts, part.contiguous())
-    stacked = torch.stack(parts)  # [world, m]
+    stacked = torch.stack(gather_tensors(part))  # [world, m]
     return torch.logsumexp(stacked, dim=0)
 
 
@@ -100,11 +102,13 @@ def allreduce_bitmap_or(words: torch.Tensor) -> torch.Tensor:
     if not is_initialized():
         return words
     if dist.get_backend() == "gloo":
-        dist.all_reduce(words, op=dist.ReduceOp.BOR)
+        staged, home = _staging_device(words)
+        dist.all_reduce(staged, op=dist.ReduceOp.BOR)
+        if home is not None:
+            words.copy_(staged.to(home))
         return words
+    parts = gather_tensors(words)
     world = get_world_size()
-    parts = [torch.empty_like(words) for _ in range(world)]
-    dist.all_gather(parts, words.contiguous())
     acc = parts[0]
     for r in range(1, world):
         acc = torch.bitwise_or(acc, parts[r])
@@ -112,23 +116,31 @@ def allreduce_bitmap_or(words: torch.Tensor) -> torch.Tensor:
     return words
 
 
+def _allreduce_inplace(t: torch.Tensor, op) -> torch.Tensor:
+    staged, home = _staging_device(t)
+    dist.all_reduce(staged, op=op)
+    if home is not None:
+        t.copy_(staged.to(home))
+    return t
+
+
 def allreduce_minmax(mins: torch.Tensor, maxs: torch.Tensor):
     """Cross-rank elementwise min/max of aggregate statistics (K18)."""
     if is_initialized():
-        dist.all_reduce(mins, op=dist.ReduceOp.MIN)
-        dist.all_reduce(maxs, op=dist.ReduceOp.MAX)
+        _allreduce_inplace(mins, dist.ReduceOp.MIN)
+        _allreduce_inplace(maxs, dist.ReduceOp.MAX)
     return mins, maxs
 
 
 def allreduce_min(t: torch.Tensor) -> torch.Tensor:
     if is_initialized():
-        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        _allreduce_inplace(t, dist.ReduceOp.MIN)
     return t
 
 
 def allreduce_max(t: torch.Tensor) -> torch.Tensor:
     if is_initialized():
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        _allreduce_inplace(t, dist.ReduceOp.MAX)
     return t
 
 
@@ -138,12 +150,11 @@ def allreduce_welford(count: float, mean: torch.Tensor, m2: torch.Tensor):
         return count, mean, m2
     world = get_world_size()
     dev = mean.device
-    counts = [torch.zeros(1, dtype=torch.float64, device=dev) for _ in range(world)]
-    means = [torch.empty_like(mean) for _ in range(world)]
-    m2s = [torch.empty_like(m2) for _ in range(world)]
-    dist.all_gather(counts, torch.tensor([count], dtype=torch.float64, device=dev))
-    dist.all_gather(means, mean.contiguous())
-    dist.all_gather(m2s, m2.contiguous())
+    counts = gather_tensors(
+        torch.tensor([count], dtype=torch.float64, device=dev)
+    )
+    means = gather_tensors(mean)
+    m2s = gather_tensors(m2)
     tot_c, tot_mean, tot_m2 = float(counts[0].item()), means[0], m2s[0]
     for r in range(1, world):
         c = float(counts[r].item())
