@@ -196,12 +196,18 @@ class GraphedExtractor:
             xb = torch.from_numpy(np.ascontiguousarray(xb))
         if xb.dtype == torch.float64:
             xb = xb.float()
-        if self.fused is not None and self.graph is not None:
-            if xb.shape[0] == self.batch and xb.dim() == 4:
-                # fused path consumes NHWC fp32
-                nhwc = (
-                    xb if xb.shape[-1] == 3 else xb.permute(0, 2, 3, 1)
-                ).contiguous()
+        if self.fused is not None:
+            # NOTE: the fused path emits the stage-3 tap flattened in NHWC
+            # order ([B, 8*8*64]); the eager torch flatten would be NCHW. A
+            # fixed feature permutation is invisible to every consumer of SA
+            # taps (L2 distances, KDE, covariances are permutation-
+            # equivariant) but mixing layouts within one dataset is NOT —
+            # so remainder batches take the fused eager call, never the
+            # torch model.
+            nhwc = (
+                xb if xb.shape[-1] == 3 else xb.permute(0, 2, 3, 1)
+            ).contiguous()
+            if self.graph is not None and xb.shape[0] == self.batch:
                 self.fstatic_x.copy_(nhwc.to(self.device, torch.float32))
                 self.graph.replay()
                 return (
@@ -209,11 +215,6 @@ class GraphedExtractor:
                      for t in self.static_taps],
                     self.static_probs.clone(),
                 )
-        elif self.fused is not None:
-            nhwc = (
-                xb if xb.dim() == 4 and xb.shape[-1] == 3
-                else xb.permute(0, 2, 3, 1)
-            ).contiguous()
             ats, logits = self.fused.forward_nhwc(
                 nhwc.to(self.device, torch.float32)
             )

@@ -46,6 +46,43 @@ TIP_DEV int swz(int u) {
   return u ^ (((u >> 4) * TIP_SWZ_M) & 15);
 }
 
+// LDS image layout. Round-1 PMC runs measured residual bank conflicts on
+// the per-tap A-fragment gathers (profiles/r01_optimization_ladder.md) and
+// sketched a "half-channel plane" layout (C/8 planes of one 16-B unit per
+// pixel, making same-row gathers stride-1) as the fix. MEASURED IN ROUND 2
+// AND REJECTED: on hardware the half-plane layout RAISES
+// SQ_LDS_BANK_CONFLICT 3.2x (1.68e7 -> 5.45e7 per 3-launch probe) and
+// costs 8% (382 -> 353 TF on resblock<32,32,16>) — the b128 service
+// grouping evidently pairs lanes across k-groups, where the packed
+// layout's (tap, half-channel) offsets already land in distinct bank
+// groups and the stride-C/8 pixel walk does not collide the way a
+// same-row-lanes model predicts. The packed unit-row-major image
+// (TIP_HALFPLANE=0) stays the default; the half-plane variant is kept
+// compilable for future counter runs.
+#ifndef TIP_HALFPLANE
+#define TIP_HALFPLANE 0
+#endif
+
+template <int H, int W, int C>
+struct Img {
+  static constexpr int UPP = C / 8;  // 16-B units per pixel
+#if TIP_HALFPLANE
+  static constexpr int RS1 = W + 2;          // units per halo row per plane
+  static constexpr int PLANE = (H + 2) * RS1;
+  static constexpr int UNITS = UPP * PLANE;
+  static constexpr int XSTEP = 1;            // unit step per +1 output x
+  static TIP_DEV int unit(int y, int x, int h) {
+    return h * PLANE + y * RS1 + x;
+  }
+#else
+  static constexpr int UNITS = (H + 2) * (W + 2) * UPP;
+  static constexpr int XSTEP = UPP;
+  static TIP_DEV int unit(int y, int x, int h) {
+    return (y * (W + 2) + x) * UPP + h;
+  }
+#endif
+};
+
 // Load a 16-B unit (8 bf16) from the swizzled LDS image.
 TIP_DEV short8 lds_read_unit(const short* lds, int u) {
   return *reinterpret_cast<const short8*>(lds + swz(u) * 8);
@@ -59,9 +96,10 @@ TIP_DEV void lds_write_unit(short* lds, int u, short8 v) {
 // [(H+2) x (W+2) x C] interior; halo is zeroed first. 256 threads.
 template <int H, int W, int C>
 TIP_DEV void stage_plane(short* lds, const short* __restrict__ gsrc) {
-  constexpr int UNITS_IMG = (H + 2) * (W + 2) * C / 8;
+  using L = Img<H, W, C>;
   const short8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
-  for (int u = threadIdx.x; u < UNITS_IMG; u += blockDim.x) lds_write_unit(lds, u, zero);
+  for (int u = threadIdx.x; u < L::UNITS; u += blockDim.x)
+    lds_write_unit(lds, u, zero);
   __syncthreads();
   constexpr int UNITS = H * W * C / 8;
   constexpr int UPP = C / 8;  // units per pixel
@@ -69,7 +107,7 @@ TIP_DEV void stage_plane(short* lds, const short* __restrict__ gsrc) {
     const int p = u / UPP;
     const int h = u - p * UPP;
     const int y = p / W, x = p - y * W;
-    const int lu = (((y + 1) * (W + 2)) + (x + 1)) * UPP + h;
+    const int lu = L::unit(y + 1, x + 1, h);
     short8 v = *reinterpret_cast<const short8*>(gsrc + (int64_t)u * 8);
     lds_write_unit(lds, lu, v);
   }
@@ -94,14 +132,13 @@ TIP_DEV void conv3x3(
   constexpr int NPIX = OH * OW;
   constexpr int PIX_TILES = NPIX / 16;
   constexpr int COUT_TILES = COUT / 16;
-  constexpr int UPP = C / 8;
-  constexpr int OUPP = COUT / 8;
+  using In = Img<H, W, C>;
+  using Out = Img<OH, OW, COUT>;
 
   const int lane = lane_id();
   const int wid = wave_id();
   const int j = lane & 15;        // output channel within tile
   const int g = lane >> 4;        // k-group (8 consecutive k)
-  constexpr int RS = (W + 2) * UPP;  // LDS units per halo row
 
   // cout-tile outer loop: the tile's weight fragments are preloaded into
   // registers ONCE (KSTEPS x 16 B per lane) so the MFMA loop is pure
@@ -127,32 +164,32 @@ TIP_DEV void conv3x3(
       const int ci = k0 & (C - 1);
       const int dy = (tap * 11) >> 5;
       const int dx = tap - dy * 3;
-      return dy * RS + dx * UPP + (ci >> 3);
+      return In::unit(dy, dx, ci >> 3);
     };
     auto a_base = [&](int pt) {
       const int apix = pt * 16 + j;
       const int aoy = apix / OW, aox = apix - aoy * OW;
-      return (aoy * STRIDE) * RS + (aox * STRIDE) * UPP;
+      return In::unit(aoy * STRIDE, aox * STRIDE, 0);
     };
     auto epilogue = [&](int pt, const f32x4& acc) {
       // the 4 regs of a C/D fragment are 4 consecutive pixels in ONE output
       // row (4 <= OW always), so the halo-image unit address is linear in
-      // reg: one div/mod per fragment, +OUPP per register
+      // reg: one div/mod per fragment, +XSTEP per register
       const int pix0 = pt * 16 + g * 4;
       const int oy = pix0 / OW, ox = pix0 - oy * OW;
-      const int u0 = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
+      const int u0 = Out::unit(oy + 1, ox + 1, cout >> 3);
       const int e = cout & 7;
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         float v = acc[reg] + bs;
         if (RESID) {
-          const short* runit = rlds + swz(u0 + reg * OUPP) * 8;
+          const short* runit = rlds + swz(u0 + reg * Out::XSTEP) * 8;
           v += __bfloat162float(reinterpret_cast<const bf16*>(runit)[e]);
         }
         v = fmaxf(v, 0.f);
         const bf16 ov = __float2bfloat16(v);
         if (TO_LDS) {
-          short* unit = out_lds + swz(u0 + reg * OUPP) * 8;
+          short* unit = out_lds + swz(u0 + reg * Out::XSTEP) * 8;
           reinterpret_cast<bf16*>(unit)[e] = ov;
         } else {
           reinterpret_cast<bf16*>(gout)[(int64_t)(pix0 + reg) * COUT + cout] = ov;
@@ -218,8 +255,8 @@ TIP_DEV void shortcut1x1_s2(
   constexpr int NPIX = OH * OW;
   constexpr int PIX_TILES = NPIX / 16;
   constexpr int COUT_TILES = COUT / 16;
-  constexpr int UPP = C / 8;
-  constexpr int OUPP = COUT / 8;
+  using In = Img<H, W, C>;
+  using Out = Img<OH, OW, COUT>;
   const int lane = lane_id();
   const int wid = wave_id();
   const int j = lane & 15;
@@ -237,8 +274,7 @@ TIP_DEV void shortcut1x1_s2(
         const int pix = p0 + j;
         const int oy = pix / OW, ox = pix - oy * OW;
         const int iy = oy * 2 + 1, ix = ox * 2 + 1;  // center tap, halo coords
-        const int u = (iy * (W + 2) + ix) * UPP + (k0 >> 3);
-        a = lds_read_unit(in_lds, u);
+        a = lds_read_unit(in_lds, In::unit(iy, ix, k0 >> 3));
       }
       const short8 b = *reinterpret_cast<const short8*>(wp + (int64_t)ks * 64 * 8);
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
@@ -247,10 +283,10 @@ TIP_DEV void shortcut1x1_s2(
     const float bs = bias[cout];
     const int pix0 = p0 + g * 4;
     const int oy = pix0 / OW, ox = pix0 - oy * OW;
-    const int u0 = ((oy + 1) * (OW + 2) + (ox + 1)) * OUPP + (cout >> 3);
+    const int u0 = Out::unit(oy + 1, ox + 1, cout >> 3);
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
-      short* unit = rlds + swz(u0 + reg * OUPP) * 8;
+      short* unit = rlds + swz(u0 + reg * Out::XSTEP) * 8;
       reinterpret_cast<bf16*>(unit)[cout & 7] = __float2bfloat16(acc[reg] + bs);
     }
   }
@@ -267,15 +303,14 @@ __launch_bounds__(256) __global__ void resblock_kernel(
     const short* __restrict__ w1, const float* __restrict__ b1,
     const short* __restrict__ w2, const float* __restrict__ b2) {
   extern __shared__ short lds[];
-  short* bufX = lds;                                // [(H+2)(W+2)C]
-  short* bufH = lds + (H + 2) * (W + 2) * C;        // [(H+2)(W+2)C]
+  short* bufX = lds;                                // image of Img<H,W,C>
+  short* bufH = lds + Img<H, W, C>::UNITS * 8;      // second image
   const int64_t img_off = (int64_t)blockIdx.x * H * W * C;
   stage_plane<H, W, C>(bufX, gin + img_off);
   // zero bufH halo (stage_plane zeroes everything first; emulate)
   {
-    constexpr int UNITS_IMG = (H + 2) * (W + 2) * C / 8;
     const short8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int u = threadIdx.x; u < UNITS_IMG; u += blockDim.x)
+    for (int u = threadIdx.x; u < Img<H, W, C>::UNITS; u += blockDim.x)
       lds_write_unit(bufH, u, zero);
   }
   __syncthreads();
@@ -296,16 +331,15 @@ __launch_bounds__(256) __global__ void downblock_kernel(
     const short* __restrict__ wsc, const float* __restrict__ bsc) {
   constexpr int OH = H / 2, OW = W / 2, C2 = 2 * C;
   extern __shared__ short lds[];
-  short* bufX = lds;                                  // [(H+2)(W+2)C]
-  short* bufH = bufX + (H + 2) * (W + 2) * C;         // [(OH+2)(OW+2)C2]
-  short* bufR = bufH + (OH + 2) * (OW + 2) * C2;      // [(OH+2)(OW+2)C2]
+  short* bufX = lds;                                   // Img<H,W,C>
+  short* bufH = bufX + Img<H, W, C>::UNITS * 8;        // Img<OH,OW,C2>
+  short* bufR = bufH + Img<OH, OW, C2>::UNITS * 8;     // Img<OH,OW,C2>
   const int64_t in_off = (int64_t)blockIdx.x * H * W * C;
   const int64_t out_off = (int64_t)blockIdx.x * OH * OW * C2;
   stage_plane<H, W, C>(bufX, gin + in_off);
   {
-    constexpr int UNITS = ((OH + 2) * (OW + 2) * C2 / 8);
     const short8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int u = threadIdx.x; u < UNITS; u += blockDim.x) {
+    for (int u = threadIdx.x; u < Img<OH, OW, C2>::UNITS; u += blockDim.x) {
       lds_write_unit(bufH, u, zero);
       lds_write_unit(bufR, u, zero);
     }
@@ -360,7 +394,7 @@ template <int H, int W, int C>
 static void resblock(int batch, const short* gin, short* gout, const short* w1,
                      const float* b1, const short* w2, const float* b2,
                      hipStream_t s) {
-  const int lds_bytes = 2 * (H + 2) * (W + 2) * C * 2;
+  const int lds_bytes = 2 * Img<H, W, C>::UNITS * 16;
   auto k = resblock_kernel<H, W, C>;
   hipFuncSetAttribute((const void*)k,
                       hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
@@ -382,7 +416,7 @@ static void downblock(int batch, const short* gin, short* gout,
                       hipStream_t s) {
   constexpr int OH = H / 2, OW = W / 2, C2 = 2 * C;
   const int lds_bytes =
-      ((H + 2) * (W + 2) * C + 2 * (OH + 2) * (OW + 2) * C2) * 2;
+      (Img<H, W, C>::UNITS + 2 * Img<OH, OW, C2>::UNITS) * 16;
   auto k = downblock_kernel<H, W, C>;
   hipFuncSetAttribute((const void*)k,
                       hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
@@ -402,7 +436,7 @@ void launch_downblock(int variant, int batch, const short* gin, short* gout,
 void launch_stem(int batch, const short* gin, short* gout, const short* w,
                  const float* b, hipStream_t s) {
   constexpr int H = 32, W = 32, CIN = 8, COUT = 16;
-  const int lds_bytes = (H + 2) * (W + 2) * CIN * 2;
+  const int lds_bytes = Img<H, W, CIN>::UNITS * 16;
   auto k = stem_kernel<H, W, CIN, COUT>;
   hipFuncSetAttribute((const void*)k,
                       hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
