@@ -256,7 +256,7 @@ __launch_bounds__(256, 2) __global__ void grouped_pairwise_kernel(
     const float* __restrict__ bnorm,  // [Ntot]
     const int* __restrict__ tseg,     // [C+1] padded test offsets (x128)
     const int* __restrict__ nseg,     // [C+1] train offsets
-    int nclasses, int Bp, int K,
+    int nclasses, int Bp, int K, int jb_max_g,
     float* __restrict__ pmin_val,     // [jb_max, Bp]
     int* __restrict__ pmin_idx,
     float2* __restrict__ pkde) {
@@ -267,16 +267,47 @@ __launch_bounds__(256, 2) __global__ void grouped_pairwise_kernel(
   __shared__ int red_i[BM][2];
   __shared__ float red_s[BM][2];
 
-  const int bi = blockIdx.y;
+  // Block->tile mapping. TIP_GXY selects the variant (hardware-swept):
+  //  0: bi = blockIdx.y, bj = blockIdx.x (column block fast — baseline)
+  //  1: grid swapped, row block fast: consecutive ids share a column tile
+  //  2: XCD-aware (1-D grid padded to 8*per): consecutive block ids land on
+  //     different XCDs (round-robin dispatch), so give each XCD a
+  //     CONTIGUOUS tile range decomposed row-block-fast — every XCD sweeps
+  //     the row blocks of one (or few) column tiles and its 2-MB B tile
+  //     stays L2-resident while the A tiles stream.
+#ifndef TIP_GXY
+#define TIP_GXY 0
+#endif
+  int bi, bj;
+#if TIP_GXY == 0
+  bi = blockIdx.y;
+  bj = blockIdx.x;
+#elif TIP_GXY == 1
+  bi = blockIdx.x;
+  bj = blockIdx.y;
+#else
+  {
+    // 1-D grid of 8*per blocks (per = ceil(nrows*jb_max/8))
+    const int nrows = (Bp + BM - 1) / BM;
+    const int total = nrows * jb_max_g;
+    constexpr int NXCD = 8;
+    const int per = gridDim.x / NXCD;
+    const int t0 = blockIdx.x;
+    const int t = (t0 % NXCD) * per + t0 / NXCD;
+    if (t >= total) return;
+    bi = t % nrows;
+    bj = t / nrows;
+  }
+#endif
   const int row0 = bi * BM;
+  if (row0 >= Bp) return;
   // class of this 128-aligned row block
   int cls = 0;
   for (int c = 0; c < nclasses; ++c)
     if (tseg[c] <= row0) cls = c;
   const int ncol0 = nseg[cls], ncol1 = nseg[cls + 1];
-  const int col0 = ncol0 + blockIdx.x * BN;
+  const int col0 = ncol0 + bj * BN;
   if (col0 >= ncol1) return;  // this class has fewer column blocks
-  const int bj = blockIdx.x;
 
   const int lane = lane_id();
   const int wid = wave_id();
@@ -513,14 +544,25 @@ void launch_pairwise_kde(const float* a, const float* b, const float* an,
   kde_combine_kernel<<<ceil_div(m, 256), 256, 0, s>>>(pkde, jb, m, out_lse);
 }
 
+// Grid geometry matching the kernel's TIP_GXY block->tile mapping.
+static dim3 grouped_grid(int bp, int jb_max) {
+#if TIP_GXY == 0
+  return dim3(jb_max, ceil_div(bp, BM));
+#elif TIP_GXY == 1
+  return dim3(ceil_div(bp, BM), jb_max);
+#else
+  const int total = ceil_div(bp, BM) * jb_max;
+  return dim3(8 * ceil_div(total, 8), 1);
+#endif
+}
+
 void launch_grouped_rowmin(const float* a, const float* b, const float* an,
                            const float* bn, const int* tseg, const int* nseg,
                            int nclasses, int bp, int k, int jb_max,
                            float* pval, int* pidx, float* out_dist,
                            int64_t* out_idx, hipStream_t s) {
-  dim3 grid(jb_max, ceil_div(bp, BM));
-  grouped_pairwise_kernel<EPI_ROWMIN><<<grid, 256, 0, s>>>(
-      a, b, an, bn, tseg, nseg, nclasses, bp, k, pval, pidx, nullptr);
+  grouped_pairwise_kernel<EPI_ROWMIN><<<grouped_grid(bp, jb_max), 256, 0, s>>>(
+      a, b, an, bn, tseg, nseg, nclasses, bp, k, jb_max, pval, pidx, nullptr);
   grouped_rowmin_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
       pval, pidx, tseg, nseg, nclasses, bp, out_dist, out_idx);
 }
@@ -529,9 +571,9 @@ void launch_grouped_kde(const float* a, const float* b, const float* an,
                         const float* bn, const int* tseg, const int* nseg,
                         int nclasses, int bp, int k, int jb_max, float2* pkde,
                         float* out_lse, hipStream_t s) {
-  dim3 grid(jb_max, ceil_div(bp, BM));
-  grouped_pairwise_kernel<EPI_KDE><<<grid, 256, 0, s>>>(
-      a, b, an, bn, tseg, nseg, nclasses, bp, k, nullptr, nullptr, pkde);
+  grouped_pairwise_kernel<EPI_KDE><<<grouped_grid(bp, jb_max), 256, 0, s>>>(
+      a, b, an, bn, tseg, nseg, nclasses, bp, k, jb_max, nullptr, nullptr,
+      pkde);
   grouped_kde_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
       pkde, tseg, nseg, nclasses, bp, out_lse);
 }
