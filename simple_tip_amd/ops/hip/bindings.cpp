@@ -255,27 +255,50 @@ torch::Tensor cam_greedy(torch::Tensor words, int64_t nbits) {
   const int nblocks = (rows + wpb - 1) / wpb;
   auto part_val = torch::empty({nblocks}, opts);
   auto part_idx = torch::empty({nblocks}, opts.dtype(torch::kInt32));
-  auto result = torch::empty({2}, opts);
+  // The greedy loop can run thousands of iterations (one per newly covered
+  // section); a host sync per iteration made it launch-latency-bound
+  // (measured 470 ms for 20k x 1000-bit profiles). Iterations are
+  // IDEMPOTENT once coverage is exhausted (the pick kernel writes -1 and
+  // leaves state untouched), so we enqueue CHAIN iterations back-to-back
+  // and sync once per chain: per-iteration cost collapses to the two
+  // kernel dispatches.
+  constexpr int CHAIN = 256;
+  auto result = torch::empty({2 * CHAIN}, opts);
   std::vector<int64_t> order;
   order.reserve(std::min<int64_t>(rows, nbits));
   auto stream = cur_stream();
-  long long host_result[2];
-  for (;;) {
-    launch_cam_iteration(
-        reinterpret_cast<const unsigned long long*>(words.data_ptr<int64_t>()),
-        rows, W,
-        reinterpret_cast<unsigned long long*>(uncovered.data_ptr<int64_t>()),
-        used.data_ptr<uint8_t>(),
-        reinterpret_cast<long long*>(part_val.data_ptr<int64_t>()),
-        part_idx.data_ptr<int>(),
-        reinterpret_cast<long long*>(result.data_ptr<int64_t>()), stream);
-    C10_HIP_CHECK(hipMemcpyAsync(host_result, result.data_ptr<int64_t>(),
-                                 2 * sizeof(long long), hipMemcpyDeviceToHost,
-                                 stream));
+  std::vector<long long> host_result(2 * CHAIN);
+  bool done = false;
+  while (!done) {
+    for (int j = 0; j < CHAIN; ++j) {
+      launch_cam_iteration(
+          reinterpret_cast<const unsigned long long*>(
+              words.data_ptr<int64_t>()),
+          rows, W,
+          reinterpret_cast<unsigned long long*>(
+              uncovered.data_ptr<int64_t>()),
+          used.data_ptr<uint8_t>(),
+          reinterpret_cast<long long*>(part_val.data_ptr<int64_t>()),
+          part_idx.data_ptr<int>(),
+          reinterpret_cast<long long*>(result.data_ptr<int64_t>()) + 2 * j,
+          stream);
+    }
+    C10_HIP_CHECK(hipMemcpyAsync(host_result.data(),
+                                 result.data_ptr<int64_t>(),
+                                 2 * CHAIN * sizeof(long long),
+                                 hipMemcpyDeviceToHost, stream));
     C10_HIP_CHECK(hipStreamSynchronize(stream));
-    if (host_result[0] < 0) break;
-    order.push_back(host_result[0]);
-    if ((int64_t)order.size() >= rows) break;
+    for (int j = 0; j < CHAIN; ++j) {
+      if (host_result[2 * j] < 0) {
+        done = true;
+        break;
+      }
+      order.push_back(host_result[2 * j]);
+      if ((int64_t)order.size() >= rows) {
+        done = true;
+        break;
+      }
+    }
   }
   return torch::tensor(order, torch::dtype(torch::kInt64));
 }
