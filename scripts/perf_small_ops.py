@@ -67,12 +67,41 @@ def main():
     timed("popcount_rows", lambda: ops.popcount_rows(prof.words))
 
     scores = prof.popcount().float()
+    ops.cam_order(scores, prof.words, prof.nbits)  # warm torch sort modules
+    torch.cuda.synchronize()
     t0 = time.perf_counter()
     order = ops.cam_order(scores, prof.words, prof.nbits)
     torch.cuda.synchronize()
-    print(f"{'cam_order (1 run)':24s} {(time.perf_counter()-t0)*1000:8.3f} ms "
+    print(f"{'cam_order (warmed)':24s} {(time.perf_counter()-t0)*1000:8.3f} ms "
           f"({int(order.shape[0])} rows)")
 
 
 if __name__ == "__main__":
     main()
+
+
+def cam_diag():
+    """Pick-count/time scaling of cam_greedy itself."""
+    from simple_tip_amd.ops import _load_compiled
+
+    ext = _load_compiled()
+    for nbits, density in ((1000, 1), (1000, 8), (8000, 1)):
+        vals = torch.rand(N, device=DEV).double()
+        thr = torch.linspace(0, 1, nbits + 1, dtype=torch.float64).to(DEV)
+        words = ops.bucketize_profile(vals, thr)
+        if density > 1:  # OR together several shifted profiles
+            w = words.clone()
+            for s in range(1, density):
+                w = w | words.roll(s * 37, dims=0)
+            words = w
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        picked = ext.cam_greedy(words.contiguous(), nbits)
+        torch.cuda.synchronize()
+        ms = (time.perf_counter() - t0) * 1000
+        print(f"cam_greedy nbits={nbits} density~{density}: {ms:8.2f} ms, "
+              f"picks={picked.numel()}, ms/pick={ms/max(1,picked.numel()):.4f}")
+
+
+if os.environ.get("TIP_CAM_DIAG") == "1":
+    cam_diag()

@@ -40,6 +40,13 @@ void launch_bucketize(const double*, const double*, int, int, int,
 void launch_cam_iteration(const unsigned long long*, int, int,
                           unsigned long long*, unsigned char*, long long*,
                           int*, long long*, hipStream_t);
+int launch_cam_greedy_block(const unsigned long long*, int, int,
+                            unsigned long long, unsigned char*, long long*,
+                            int*, hipStream_t);
+void launch_cam_greedy_coop(const unsigned long long*, int, int,
+                            unsigned long long, unsigned char*,
+                            unsigned long long*, long long*, int*, long long*,
+                            int*, int*, int*, hipStream_t);
 void launch_softmax_scores(const float*, int, int, float*, float*, float*,
                            float*, hipStream_t);
 void launch_mfma_probe(const short*, const short*, float*, hipStream_t);
@@ -247,26 +254,53 @@ torch::Tensor cam_greedy(torch::Tensor words, int64_t nbits) {
               words.dim() == 2 && words.is_contiguous());
   const int rows = words.size(0), W = words.size(1);
   auto opts = words.options();
-  auto uncovered = torch::full({W}, -1, opts);  // all ones
   const int tail = nbits % 64;
-  if (tail) uncovered.index_put_({W - 1}, (int64_t)((1ull << tail) - 1));
+  const unsigned long long tail_mask =
+      tail ? ((1ull << tail) - 1) : ~0ull;
   auto used = torch::zeros({rows}, opts.dtype(torch::kUInt8));
+  auto stream = cur_stream();
+
+  // Fast path: the WHOLE greedy loop in one persistent cooperative kernel
+  // (64 resident blocks + software grid barriers). The per-iteration
+  // two-kernel form paid ~200 us of dependent-launch latency per pick and
+  // a single-block loop is sweep-bound on one CU (~340 us/pick); this form
+  // has zero dispatch and a grid-parallel sweep (profiles/r02).
+  {
+    auto order_dev = torch::empty({rows}, opts);
+    auto n_dev = torch::zeros({1}, opts.dtype(torch::kInt32));
+    auto uncovered_dev = torch::empty({W}, opts);
+    auto part_val = torch::empty({64}, opts);
+    auto part_idx = torch::empty({64}, opts.dtype(torch::kInt32));
+    auto aux = torch::zeros({2}, opts.dtype(torch::kInt32));  // barrier, win
+    launch_cam_greedy_coop(
+        reinterpret_cast<const unsigned long long*>(
+            words.data_ptr<int64_t>()),
+        rows, W, tail_mask, used.data_ptr<uint8_t>(),
+        reinterpret_cast<unsigned long long*>(
+            uncovered_dev.data_ptr<int64_t>()),
+        reinterpret_cast<long long*>(part_val.data_ptr<int64_t>()),
+        part_idx.data_ptr<int>(),
+        reinterpret_cast<long long*>(order_dev.data_ptr<int64_t>()),
+        n_dev.data_ptr<int>(), aux.data_ptr<int>(),
+        aux.data_ptr<int>() + 1, stream);
+    const int n = n_dev.cpu().item<int>();
+    return order_dev.narrow(0, 0, n).cpu();
+  }
+
+  // Fallback (coverage mask too wide for LDS): per-iteration kernels with
+  // CHAIN iterations enqueued per host sync (iterations are idempotent
+  // once coverage is exhausted, so over-running is safe).
+  auto uncovered = torch::full({W}, -1, opts);  // all ones
+  if (tail)
+    uncovered.index_put_({W - 1}, (int64_t)tail_mask);
   const int wpb = 8;
   const int nblocks = (rows + wpb - 1) / wpb;
   auto part_val = torch::empty({nblocks}, opts);
   auto part_idx = torch::empty({nblocks}, opts.dtype(torch::kInt32));
-  // The greedy loop can run thousands of iterations (one per newly covered
-  // section); a host sync per iteration made it launch-latency-bound
-  // (measured 470 ms for 20k x 1000-bit profiles). Iterations are
-  // IDEMPOTENT once coverage is exhausted (the pick kernel writes -1 and
-  // leaves state untouched), so we enqueue CHAIN iterations back-to-back
-  // and sync once per chain: per-iteration cost collapses to the two
-  // kernel dispatches.
   constexpr int CHAIN = 256;
   auto result = torch::empty({2 * CHAIN}, opts);
   std::vector<int64_t> order;
   order.reserve(std::min<int64_t>(rows, nbits));
-  auto stream = cur_stream();
   std::vector<long long> host_result(2 * CHAIN);
   bool done = false;
   while (!done) {
