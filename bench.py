@@ -353,8 +353,65 @@ def main():
             pass
 
     phase_log = os.environ.get("TIP_BENCH_PHASES") == "1"
+    seen = torch.tensor(sorted(train_classes), device=device)
+
+    # ---- two-stage software pipeline (GPU): forward[i+1] runs on its own
+    # stream while the scoring of batch i (softmax family + grouped DSA/LSA)
+    # runs on the main stream. The extractor's hipGraph replay writes static
+    # output buffers, so each forward's (ats, probs) are copied into parity
+    # double-buffers before the next replay may clobber them; per-parity
+    # events order buffer reuse. Scoring must not host-sync (the class-remap
+    # guard is branch-free) or the overlap collapses.
+    fwd_stream = torch.cuda.Stream() if on_gpu else None
+    fwd_done = [torch.cuda.Event(), torch.cuda.Event()] if on_gpu else None
+    score_done = [torch.cuda.Event(), torch.cuda.Event()] if on_gpu else None
+    out_bufs = [None, None]
+
+    def launch_forward(i):
+        """Enqueue batch i's extraction on the forward stream."""
+        p = i % 2
+        with torch.cuda.stream(fwd_stream):
+            fwd_stream.wait_event(score_done[p])  # buffer p free?
+            x = get_batch(i)
+            ats, probs = extractor(x)
+            mark_consumed(i)
+            prefetch(i + 1)
+            if out_bufs[p] is None:
+                out_bufs[p] = (torch.empty_like(ats), torch.empty_like(probs))
+            out_bufs[p][0].copy_(ats)
+            out_bufs[p][1].copy_(probs)
+            fwd_done[p].record(fwd_stream)
+
+    def score(i):
+        """Score batch i's (already extracted) activations on this stream."""
+        p = i % 2
+        s = torch.cuda.current_stream()
+        s.wait_event(fwd_done[p])
+        ats, probs = out_bufs[p]
+        # branch-free synthetic-data guard: remap predictions of classes
+        # unseen in training (no host sync — it would serialise the pipe)
+        pred = probs.argmax(dim=1)
+        pred = torch.where(torch.isin(pred, seen), pred, seen[0])
+        unc = ops.softmax_uncertainties(probs)
+        if fused_prio is not None and fused_prio.lsa_ready:
+            dsa_scores, lsa_scores = fused_prio(ats, pred)
+        else:
+            dsa_scores = dsa(ats, pred)
+            lsa_scores = lsa(ats, pred)
+        gini = unc["deep_gini"]
+        # publish score shards (tiny, latency-bound on xGMI). In shard-train
+        # mode the DSA/LSA partials already merged inside the prioritizer and
+        # every rank holds full-batch scores — nothing left to gather.
+        if world > 1 and not shard_train:
+            n_total = args.batch * world
+            gini_all = pdist.allgather_rows(gini, n_total)
+            dsa_all = pdist.allgather_rows(dsa_scores.float().to(device), n_total)
+            _ = (gini_all, dsa_all)
+        score_done[p].record(torch.cuda.current_stream())
+        return gini, dsa_scores, lsa_scores, pred
 
     def step(i, timed_phases=False):
+        """Non-pipelined step (CPU path and phase-instrumented runs)."""
         marks = []
 
         def mark(name):
@@ -366,14 +423,9 @@ def main():
         mark("start")
         ats, probs = extractor(x)
         mark_consumed(i)
-        prefetch(i + 1)  # overlap next batch's H2D with this step's compute
+        prefetch(i + 1)
         pred = probs.argmax(dim=1)
-        # synthetic-data guard: random-init models may emit a class absent
-        # from the train predictions; remap to a seen class
-        seen = torch.tensor(sorted(train_classes), device=pred.device)
-        ok = torch.isin(pred, seen)
-        if not bool(ok.all()):
-            pred = torch.where(ok, pred, seen[0])
+        pred = torch.where(torch.isin(pred, seen.to(pred.device)), pred, seen.to(pred.device)[0])
         mark("forward")
         unc = ops.softmax_uncertainties(probs)
         mark("unc")
@@ -386,9 +438,6 @@ def main():
             lsa_scores = lsa(ats, pred)
             mark("lsa")
         gini = unc["deep_gini"]
-        # publish score shards (tiny, latency-bound on xGMI). In shard-train
-        # mode the DSA/LSA partials already merged inside the prioritizer and
-        # every rank holds full-batch scores — nothing left to gather.
         if world > 1 and not shard_train:
             n_total = args.batch * world
             gini_all = pdist.allgather_rows(gini, n_total)
@@ -403,9 +452,18 @@ def main():
             log(rank, "phases: " + " ".join(parts))
         return gini, dsa_scores, lsa_scores, pred
 
-    log(rank, f"warmup x{args.warmup}")
-    for i in range(args.warmup):
-        step(i, timed_phases=phase_log and on_gpu and i == args.warmup - 1)
+    pipelined = on_gpu and os.environ.get("TIP_NO_PIPELINE") != "1"
+    log(rank, f"warmup x{args.warmup} (pipelined={pipelined})")
+    if pipelined:
+        for i in range(args.warmup):
+            launch_forward(i)
+            step_out = score(i)
+        if phase_log:
+            torch.cuda.synchronize()
+            step(args.warmup - 1, timed_phases=True)  # phase breakdown, unpipelined
+    else:
+        for i in range(args.warmup):
+            step(i, timed_phases=phase_log and on_gpu and i == args.warmup - 1)
 
     pdist.barrier()
     if on_gpu:
@@ -413,7 +471,14 @@ def main():
     t0 = time.perf_counter()
     last = None
     for i in range(args.steps):
-        last = step(args.warmup + i)
+        if pipelined:
+            # enqueue batch i's forward, then its scoring: the host runs
+            # ahead, so forward[i+1] lands on the forward stream while the
+            # GPU still executes scoring[i] on the main stream
+            launch_forward(args.warmup + i)
+            last = score(args.warmup + i)
+        else:
+            last = step(args.warmup + i)
     pdist.barrier()
     if on_gpu:
         torch.cuda.synchronize()
