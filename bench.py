@@ -284,8 +284,15 @@ def main():
         try:
             from simple_tip_amd.engine.serving import FusedPrioritizer
 
-            fused_prio = FusedPrioritizer(dsa, lsa, device)
-            log(rank, f"fused prioritizer ready (lsa={fused_prio.lsa_ready})")
+            # bf16 pairwise kernels by default (the bench's declared compute
+            # dtype is bf16; TIP_FP32_PAIRWISE=1 restores the fp32 MFMA path)
+            pdtype = (
+                None if os.environ.get("TIP_FP32_PAIRWISE") == "1"
+                else torch.bfloat16
+            )
+            fused_prio = FusedPrioritizer(dsa, lsa, device, pairwise_dtype=pdtype)
+            log(rank, f"fused prioritizer ready (lsa={fused_prio.lsa_ready}, "
+                      f"bf16={fused_prio.bf16})")
         except Exception as e:  # pragma: no cover
             log(rank, f"fused prioritizer unavailable: {e!r}")
 

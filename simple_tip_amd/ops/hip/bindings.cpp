@@ -28,6 +28,13 @@ void launch_grouped_rowmin(const float*, const float*, const float*,
 void launch_grouped_kde(const float*, const float*, const float*,
                         const float*, const int*, const int*, int, int, int,
                         int, float2*, float*, hipStream_t);
+void launch_grouped_rowmin_bf16(const short*, const short*, const float*,
+                                const float*, const int*, const int*, int,
+                                int, int, int, float*, int*, float*,
+                                int64_t*, hipStream_t);
+void launch_grouped_kde_bf16(const short*, const short*, const float*,
+                             const float*, const int*, const int*, int, int,
+                             int, int, float2*, float*, hipStream_t);
 void launch_profile(int, const float*, const unsigned char*, const float*,
                     const float*, float, int, int, int, int,
                     unsigned long long*, long long*, hipStream_t);
@@ -152,6 +159,52 @@ torch::Tensor grouped_kde(torch::Tensor testWS, torch::Tensor trainWS,
   launch_grouped_kde(
       testWS.data_ptr<float>(), trainWS.data_ptr<float>(),
       an.data_ptr<float>(), bnorm.data_ptr<float>(), tseg.data_ptr<int>(),
+      nseg.data_ptr<int>(), nclasses, bp, k, jb_max,
+      reinterpret_cast<float2*>(pkde.data_ptr<float>()),
+      out.data_ptr<float>(), cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> grouped_rowmin_bf16(
+    torch::Tensor testS, torch::Tensor trainS, torch::Tensor tseg,
+    torch::Tensor nseg, torch::Tensor anorm, torch::Tensor bnorm,
+    int64_t jb_max) {
+  TORCH_CHECK(testS.is_cuda() && testS.dtype() == torch::kBFloat16 &&
+              testS.is_contiguous());
+  TORCH_CHECK(trainS.dtype() == torch::kBFloat16 && trainS.is_contiguous());
+  const int bp = testS.size(0), k = testS.size(1);
+  const int nclasses = tseg.size(0) - 1;
+  auto fopts = anorm.options();
+  auto pval = torch::empty({jb_max, bp}, fopts);
+  auto pidx = torch::empty({jb_max, bp}, fopts.dtype(torch::kInt32));
+  auto dist = torch::empty({bp}, fopts);
+  auto idx = torch::empty({bp}, fopts.dtype(torch::kInt64));
+  launch_grouped_rowmin_bf16(
+      reinterpret_cast<const short*>(testS.data_ptr()),
+      reinterpret_cast<const short*>(trainS.data_ptr()),
+      anorm.data_ptr<float>(), bnorm.data_ptr<float>(), tseg.data_ptr<int>(),
+      nseg.data_ptr<int>(), nclasses, bp, k, jb_max, pval.data_ptr<float>(),
+      pidx.data_ptr<int>(), dist.data_ptr<float>(), idx.data_ptr<int64_t>(),
+      cur_stream());
+  return {dist, idx};
+}
+
+torch::Tensor grouped_kde_bf16(torch::Tensor testWS, torch::Tensor trainWS,
+                               torch::Tensor tseg, torch::Tensor nseg,
+                               torch::Tensor anorm, torch::Tensor bnorm,
+                               int64_t jb_max) {
+  TORCH_CHECK(testWS.is_cuda() && testWS.dtype() == torch::kBFloat16 &&
+              testWS.is_contiguous());
+  TORCH_CHECK(trainWS.dtype() == torch::kBFloat16 && trainWS.is_contiguous());
+  const int bp = testWS.size(0), k = testWS.size(1);
+  const int nclasses = tseg.size(0) - 1;
+  auto fopts = anorm.options();
+  auto pkde = torch::empty({jb_max, bp, 2}, fopts);
+  auto out = torch::empty({bp}, fopts);
+  launch_grouped_kde_bf16(
+      reinterpret_cast<const short*>(testWS.data_ptr()),
+      reinterpret_cast<const short*>(trainWS.data_ptr()),
+      anorm.data_ptr<float>(), bnorm.data_ptr<float>(), tseg.data_ptr<int>(),
       nseg.data_ptr<int>(), nclasses, bp, k, jb_max,
       reinterpret_cast<float2*>(pkde.data_ptr<float>()),
       out.data_ptr<float>(), cur_stream());
@@ -449,6 +502,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kde_logsumexp", &kde_logsumexp);
   m.def("grouped_rowmin", &grouped_rowmin);
   m.def("grouped_kde", &grouped_kde);
+  m.def("grouped_rowmin_bf16", &grouped_rowmin_bf16);
+  m.def("grouped_kde_bf16", &grouped_kde_bf16);
   m.def("profile", &profile);
   m.def("pack_bits", &pack_bits);
   m.def("popcount_rows", &popcount_rows);

@@ -457,6 +457,206 @@ __global__ void grouped_kde_combine_kernel(
   out_lse[i] = (ss > 0.f) ? (mm + __logf(ss)) : -FLT_MAX;
 }
 
+// ---------------------------------------------------------------------------
+// bf16 grouped variant (fp32 accumulate). The activation traces come out of
+// a bf16 forward, so bf16 operands lose almost nothing while the MFMA
+// throughput ceiling rises ~16x over v_mfma_f32_32x32x2_f32 (the fp32 path
+// measured ~95-119 TF; dense bf16 peak is ~2.5 PF). Fragment layout is the
+// one hardware-verified by mfma_probe (resnet_fused.hip): A lane l holds
+// A[i=l&15][k=(l>>4)*8+e]; B lane l holds B-col[j=l&15][same k]; D row =
+// (l>>4)*4+reg, col = l&15. LDS tiles are [128][BBK+8] halves — the +8 pad
+// makes the 16-lane fragment gathers stride 5 units (coprime to the bank
+// period). Partial outputs are identical in format to the fp32 grouped
+// kernel, so the same combine kernels run afterwards.
+// ---------------------------------------------------------------------------
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4b = __attribute__((ext_vector_type(4))) float;
+
+constexpr int BBK = 32;             // k per MFMA / staging step
+constexpr int BROW = BBK + 8;       // LDS halves per tile row (pad)
+
+// Stage a [128, 32] bf16 tile: thread t loads row r = t>>1, halves
+// kq = (t&1)*16 .. +15 (two 16-B units); OOB rows/ks stage zeros.
+struct BStage {
+  bf16x8 v[2];
+};
+
+TIP_DEV BStage bstage_load(const short* __restrict__ src, int rows, int K,
+                           int row0, int k0) {
+  BStage s;
+  const int t = threadIdx.x;
+  const int r = t >> 1;
+  const int kq = (t & 1) * 16;
+  const int grow = row0 + r;
+  const bf16x8 zero = {0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    const int gk = k0 + kq + h * 8;
+    if (grow < rows && gk + 7 < K) {
+      s.v[h] = *reinterpret_cast<const bf16x8*>(&src[(int64_t)grow * K + gk]);
+    } else if (grow < rows) {
+      bf16x8 tmp = zero;
+      for (int e = 0; e < 8; ++e)
+        if (gk + e < K) tmp[e] = src[(int64_t)grow * K + gk + e];
+      s.v[h] = tmp;
+    } else {
+      s.v[h] = zero;
+    }
+  }
+  return s;
+}
+
+TIP_DEV void bstage_write(short* lds, const BStage& s) {
+  const int t = threadIdx.x;
+  const int r = t >> 1;
+  const int kq = (t & 1) * 16;
+  *reinterpret_cast<bf16x8*>(&lds[r * BROW + kq]) = s.v[0];
+  *reinterpret_cast<bf16x8*>(&lds[r * BROW + kq + 8]) = s.v[1];
+}
+
+template <int EPI>
+__launch_bounds__(256, 2) __global__ void grouped_pairwise_bf16_kernel(
+    const short* __restrict__ A,      // [Bp, K] bf16, class-sorted padded
+    const short* __restrict__ B,      // [Ntot, K] bf16 class-concatenated
+    const float* __restrict__ anorm,  // [Bp] norms of the bf16 values
+    const float* __restrict__ bnorm,  // [Ntot]
+    const int* __restrict__ tseg, const int* __restrict__ nseg,
+    int nclasses, int Bp, int K,
+    float* __restrict__ pmin_val, int* __restrict__ pmin_idx,
+    float2* __restrict__ pkde) {
+  __shared__ short As[128 * BROW];
+  __shared__ short Bs[128 * BROW];
+  __shared__ float red_v[BM][2];
+  __shared__ int red_i[BM][2];
+  __shared__ float red_s[BM][2];
+
+  const int bi = blockIdx.y;
+  const int bj = blockIdx.x;
+  const int row0 = bi * BM;
+  int cls = 0;
+  for (int c = 0; c < nclasses; ++c)
+    if (tseg[c] <= row0) cls = c;
+  const int ncol0 = nseg[cls], ncol1 = nseg[cls + 1];
+  const int col0 = ncol0 + bj * BN;
+  if (col0 >= ncol1) return;
+
+  const int lane = lane_id();
+  const int wid = wave_id();
+  const int wr = wid >> 1;  // row half (64 rows)
+  const int wc = wid & 1;   // col half (64 cols)
+  const int fj = lane & 15;
+  const int fg = lane >> 4;
+
+  f32x4b acc[4][4] = {};
+  BStage ra = bstage_load(A, Bp, K, row0, 0);
+  BStage rb = bstage_load(B, ncol1, K, col0, 0);
+  bstage_write(As, ra);
+  bstage_write(Bs, rb);
+  __syncthreads();
+  for (int k0 = 0; k0 < K; k0 += BBK) {
+    const bool has_next = (k0 + BBK) < K;
+    if (has_next) {
+      ra = bstage_load(A, Bp, K, row0, k0 + BBK);
+      rb = bstage_load(B, ncol1, K, col0, k0 + BBK);
+    }
+    bf16x8 af[4], bf[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      const int i = wr * 64 + t * 16 + fj;
+      const int j = wc * 64 + t * 16 + fj;
+      af[t] = *reinterpret_cast<const bf16x8*>(&As[i * BROW + fg * 8]);
+      bf[t] = *reinterpret_cast<const bf16x8*>(&Bs[j * BROW + fg * 8]);
+    }
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int tc = 0; tc < 4; ++tc)
+        acc[tr][tc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[tr], bf[tc], acc[tr][tc], 0, 0, 0);
+    __syncthreads();
+    if (has_next) {
+      bstage_write(As, ra);
+      bstage_write(Bs, rb);
+    }
+    __syncthreads();
+  }
+
+  // Epilogue. D tile (tr, tc): row i = wr*64 + tr*16 + fg*4 + reg,
+  // col j = wc*64 + tc*16 + fj. Per row: reduce over this wave's 64 cols
+  // (4 tc in registers + 16 fj lanes via xor shuffles), stash per col-half.
+#pragma unroll
+  for (int tr = 0; tr < 4; ++tr) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row_local = wr * 64 + tr * 16 + fg * 4 + reg;
+      const int gi = row0 + row_local;
+      const float an = anorm[gi];
+      MinIdx mi{FLT_MAX, 0x7fffffff};
+      float dvals[4];
+#pragma unroll
+      for (int tc = 0; tc < 4; ++tc) {
+        const int gj = col0 + wc * 64 + tc * 16 + fj;
+        float d;
+        if (gj < ncol1) {
+          d = fmaxf(an + bnorm[gj] - 2.f * acc[tr][tc][reg], 0.f);
+        } else {
+          d = FLT_MAX;
+        }
+        dvals[tc] = d;
+        mi = min_idx_combine(mi, MinIdx{d, gj});
+      }
+      // 16-lane (same fg) xor reduce; offsets < 16 stay in the group
+      for (int off = 8; off >= 1; off >>= 1) {
+        MinIdx o;
+        o.v = __shfl_xor(mi.v, off);
+        o.i = __shfl_xor(mi.i, off);
+        mi = min_idx_combine(mi, o);
+      }
+      if (EPI == EPI_ROWMIN) {
+        if (fj == 0) {
+          red_v[row_local][wc] = mi.v;
+          red_i[row_local][wc] = mi.i;
+        }
+      } else {
+        const float tmax = (mi.v == FLT_MAX) ? -FLT_MAX : -0.5f * mi.v;
+        float s = 0.f;
+#pragma unroll
+        for (int tc = 0; tc < 4; ++tc)
+          if (dvals[tc] != FLT_MAX) s += __expf(-0.5f * dvals[tc] - tmax);
+        for (int off = 8; off >= 1; off >>= 1) s += __shfl_xor(s, off);
+        if (fj == 0) {
+          red_v[row_local][wc] = tmax;
+          red_s[row_local][wc] = s;
+        }
+      }
+    }
+  }
+  __syncthreads();
+  for (int r = threadIdx.x; r < BM; r += blockDim.x) {
+    const int i = row0 + r;
+    if (i >= Bp) continue;
+    if (EPI == EPI_ROWMIN) {
+      MinIdx best = min_idx_combine(
+          MinIdx{red_v[r][0], red_i[r][0]}, MinIdx{red_v[r][1], red_i[r][1]});
+      pmin_val[(int64_t)bj * Bp + i] = best.v;
+      pmin_idx[(int64_t)bj * Bp + i] = best.i;
+    } else {
+      float m0 = red_v[r][0], s0 = red_s[r][0];
+      float m1 = red_v[r][1], s1 = red_s[r][1];
+      float mm, ss;
+      if (m0 >= m1) {
+        mm = m0;
+        ss = s0 + ((s1 > 0.f) ? s1 * __expf(m1 - m0) : 0.f);
+      } else {
+        mm = m1;
+        ss = s1 + ((s0 > 0.f) ? s0 * __expf(m0 - m1) : 0.f);
+      }
+      pkde[(int64_t)bj * Bp + i] = make_float2(mm, ss);
+    }
+  }
+}
+
 // Deterministic cross-block-column combines (ascending bj keeps np.argmin
 // lowest-index tie semantics; fixed order keeps results bitwise stable).
 __global__ void rowmin_combine_kernel(
@@ -574,6 +774,31 @@ void launch_grouped_kde(const float* a, const float* b, const float* an,
   grouped_pairwise_kernel<EPI_KDE><<<grouped_grid(bp, jb_max), 256, 0, s>>>(
       a, b, an, bn, tseg, nseg, nclasses, bp, k, jb_max, nullptr, nullptr,
       pkde);
+  grouped_kde_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
+      pkde, tseg, nseg, nclasses, bp, out_lse);
+}
+
+void launch_grouped_rowmin_bf16(const short* a, const short* b,
+                                const float* an, const float* bn,
+                                const int* tseg, const int* nseg,
+                                int nclasses, int bp, int k, int jb_max,
+                                float* pval, int* pidx, float* out_dist,
+                                int64_t* out_idx, hipStream_t s) {
+  dim3 grid(jb_max, ceil_div(bp, BM));
+  grouped_pairwise_bf16_kernel<EPI_ROWMIN><<<grid, 256, 0, s>>>(
+      a, b, an, bn, tseg, nseg, nclasses, bp, k, pval, pidx, nullptr);
+  grouped_rowmin_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
+      pval, pidx, tseg, nseg, nclasses, bp, out_dist, out_idx);
+}
+
+void launch_grouped_kde_bf16(const short* a, const short* b, const float* an,
+                             const float* bn, const int* tseg,
+                             const int* nseg, int nclasses, int bp, int k,
+                             int jb_max, float2* pkde, float* out_lse,
+                             hipStream_t s) {
+  dim3 grid(jb_max, ceil_div(bp, BM));
+  grouped_pairwise_bf16_kernel<EPI_KDE><<<grid, 256, 0, s>>>(
+      a, b, an, bn, tseg, nseg, nclasses, bp, k, nullptr, nullptr, pkde);
   grouped_kde_combine_kernel<<<ceil_div(bp, 256), 256, 0, s>>>(
       pkde, tseg, nseg, nclasses, bp, out_lse);
 }
