@@ -423,7 +423,9 @@ def main():
 
         def mark(name):
             if timed_phases:
-                torch.cuda.synchronize()
+                # sync only the compute stream: a device-wide sync would
+                # also wait for the overlapped H2D prefetch and mis-bill it
+                torch.cuda.current_stream().synchronize()
                 marks.append((name, time.perf_counter()))
 
         x = get_batch(i)
