@@ -342,6 +342,84 @@ static void run_verify(int perclass_test, int perclass_train, int k) {
   }
 }
 
+// bf16 grouped rowmin at the same geometry (cross-checks vs fp32 run).
+static void run_grouped_bf16(int iters, int perclass_test, int perclass_train,
+                             int k) {
+  const int C = 10;
+  const int pc_pad = (perclass_test + 127) / 128 * 128;
+  const int bp = C * pc_pad;
+  const int ntot = C * perclass_train;
+  const int jb_max = (perclass_train + BN - 1) / BN;
+  short *a16, *b16;
+  float *an, *bn, *pval, *dist;
+  int* pidx;
+  int64_t* idx;
+  int *tseg, *nseg;
+  CHECK(hipMalloc(&a16, (size_t)bp * k * 2));
+  CHECK(hipMalloc(&b16, (size_t)ntot * k * 2));
+  CHECK(hipMalloc(&an, bp * 4));
+  CHECK(hipMalloc(&bn, ntot * 4));
+  CHECK(hipMalloc(&pval, (size_t)jb_max * bp * 4));
+  CHECK(hipMalloc(&pidx, (size_t)jb_max * bp * 4));
+  CHECK(hipMalloc(&dist, bp * 4));
+  CHECK(hipMalloc(&idx, bp * 8));
+  CHECK(hipMalloc(&tseg, (C + 1) * 4));
+  CHECK(hipMalloc(&nseg, (C + 1) * 4));
+  std::vector<int> th(C + 1), nh(C + 1);
+  for (int c = 0; c <= C; ++c) {
+    th[c] = c * pc_pad;
+    nh[c] = c * perclass_train;
+  }
+  CHECK(hipMemcpy(tseg, th.data(), (C + 1) * 4, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(nseg, nh.data(), (C + 1) * 4, hipMemcpyHostToDevice));
+  const size_t nmax = (size_t)(bp > ntot ? bp : ntot) * k;
+  std::vector<short> h16(nmax);
+  std::vector<float> norms(bp > ntot ? bp : ntot, 0.f);
+  lcg_state = 12345u;
+  for (size_t i = 0; i < nmax; ++i) {
+    const float f = frand_det();
+    unsigned u;
+    memcpy(&u, &f, 4);
+    h16[i] = (short)(u >> 16);
+    float bf;
+    u &= 0xffff0000u;
+    memcpy(&bf, &u, 4);
+    norms[i / k] += bf * bf;
+  }
+  CHECK(hipMemcpy(a16, h16.data(), (size_t)bp * k * 2, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(b16, h16.data(), (size_t)ntot * k * 2, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(an, norms.data(), bp * 4, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(bn, norms.data(), ntot * 4, hipMemcpyHostToDevice));
+  launch_grouped_rowmin_bf16(a16, b16, an, bn, tseg, nseg, C, bp, k, jb_max,
+                             pval, pidx, dist, idx, 0);
+  CHECK(hipDeviceSynchronize());
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0);
+  hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    launch_grouped_rowmin_bf16(a16, b16, an, bn, tseg, nseg, C, bp, k,
+                               jb_max, pval, pidx, dist, idx, 0);
+  hipEventRecord(t1);
+  CHECK(hipDeviceSynchronize());
+  float ms;
+  hipEventElapsedTime(&ms, t0, t1);
+  const double tf = 2.0 * bp * perclass_train * k * iters / (ms / 1e3) / 1e12;
+  std::vector<float> dh(bp);
+  std::vector<int64_t> ih(bp);
+  CHECK(hipMemcpy(dh.data(), dist, bp * 4, hipMemcpyDeviceToHost));
+  CHECK(hipMemcpy(ih.data(), idx, bp * 8, hipMemcpyDeviceToHost));
+  double dsum = 0;
+  long long isum = 0;
+  for (int i = 0; i < bp; ++i) {
+    dsum += dh[i];
+    isum += ih[i];
+  }
+  printf("grouped_rowmin_bf16 C=%d bp=%d ntr/cls=%d k=%d: %.3f ms/iter, "
+         "%.1f TF  checksum d=%.6e i=%lld\n",
+         C, bp, perclass_train, k, ms / iters, tf, dsum, isum);
+}
+
 int main(int argc, char** argv) {
   const char* which = argc > 1 ? argv[1] : "pairwise";
   const int iters = argc > 2 ? atoi(argv[2]) : 10;
@@ -351,6 +429,10 @@ int main(int argc, char** argv) {
   else if (!strcmp(which, "grouped"))
     run_grouped(iters, argc > 3 ? atoi(argv[3]) : 1024,
                 argc > 4 ? atoi(argv[4]) : 1500, argc > 5 ? atoi(argv[5]) : 4096);
+  else if (!strcmp(which, "grouped16"))
+    run_grouped_bf16(iters, argc > 3 ? atoi(argv[3]) : 1024,
+                     argc > 4 ? atoi(argv[4]) : 1500,
+                     argc > 5 ? atoi(argv[5]) : 4096);
   else if (!strcmp(which, "verify"))
     run_verify(argc > 2 ? atoi(argv[2]) : 1024, argc > 3 ? atoi(argv[3]) : 1500,
                argc > 4 ? atoi(argv[4]) : 4096);
