@@ -571,6 +571,10 @@ def main():
                         ),
                         "train_ats": args.train_n,
                         "at_width": 4096,
+                        "input_residency": (
+                            "host-pinned" if os.environ.get("TIP_HOST_POOL") == "1"
+                            else "hbm"
+                        ),
                         "scorers": "gini+softmax-family+dsa+pc-lsa",
                         "apfd_gini_lastbatch": None if np.isnan(apfd) else apfd,
                         "apfd_gini_nominal_lastbatch": (
