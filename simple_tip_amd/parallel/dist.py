@@ -27,11 +27,16 @@ def init_from_env(backend: Optional[str] = None) -> Tuple[int, int, torch.device
         dev = torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
         return 0, 1, dev
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # TIP_DIST_BACKEND=gloo lets a multi-rank run share one GPU (RCCL
+        # refuses duplicate devices) — used for in-lease validation
+        backend = os.environ.get("TIP_DIST_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo"
+        )
     local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
-        dev = torch.device(f"cuda:{local_rank}")
+        dev_id = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev_id)
+        dev = torch.device(f"cuda:{dev_id}")
     else:
         dev = torch.device("cpu")
     if not dist.is_initialized():
