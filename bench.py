@@ -237,7 +237,14 @@ def main():
         "bench_cifar10", "train", args.train_n, (3, 32, 32), 10
     )
     train_x = torch.from_numpy(base_x)
-    extractor = GraphedExtractor(model, args.batch, device, use_graph=on_gpu)
+    # in shard-train mode each rank forwards only its slice of the batch, so
+    # the hipGraph is captured at the slice size
+    if args.shard_train and world > 1:
+        _sl0 = pdist.shard_slice(args.batch)
+        cap_batch = _sl0.stop - _sl0.start
+    else:
+        cap_batch = args.batch
+    extractor = GraphedExtractor(model, cap_batch, device, use_graph=on_gpu)
 
     shard_train = bool(args.shard_train) and world > 1
     if shard_train:
@@ -387,13 +394,31 @@ def main():
     score_done = [torch.cuda.Event(), torch.cuda.Event()] if on_gpu else None
     out_bufs = [None, None]
 
+    def extract(x):
+        """AT extraction for one global batch.
+
+        In shard-train mode the FORWARD is sharded too: each rank extracts
+        its slice of the (replicated) batch and the ats/probs all-gather,
+        so every rank scores identical tensors — per-class segment sizes
+        (and therefore every later collective's shape) agree across ranks
+        BY CONSTRUCTION, independent of nondeterministic training/forward
+        differences between ranks.
+        """
+        if shard_train:
+            sl = pdist.shard_slice(x.shape[0])
+            a_l, p_l = extractor(x[sl].contiguous())
+            ats = pdist.allgather_rows(a_l.contiguous(), x.shape[0])
+            probs = pdist.allgather_rows(p_l.contiguous(), x.shape[0])
+            return ats, probs
+        return extractor(x)
+
     def launch_forward(i):
         """Enqueue batch i's extraction on the forward stream."""
         p = i % 2
         with torch.cuda.stream(fwd_stream):
             fwd_stream.wait_event(score_done[p])  # buffer p free?
             x = get_batch(i)
-            ats, probs = extractor(x)
+            ats, probs = extract(x)
             mark_consumed(i)
             prefetch(i + 1)
             if out_bufs[p] is None:
@@ -443,7 +468,7 @@ def main():
 
         x = get_batch(i)
         mark("start")
-        ats, probs = extractor(x)
+        ats, probs = extract(x)
         mark_consumed(i)
         prefetch(i + 1)
         pred = probs.argmax(dim=1)
