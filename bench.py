@@ -258,7 +258,7 @@ def main():
     at_parts, pred_parts = [], []
     for s in range(0, my_rows.shape[0], 1024):
         a, p = extractor(my_rows[s : s + 1024])
-        at_parts.append(a)
+        at_parts.append(a.float())  # SA fits are fp32 (setup, untimed)
         pred_parts.append(p.argmax(dim=1))
     train_ats = torch.cat(at_parts)
     train_pred = torch.cat(pred_parts)

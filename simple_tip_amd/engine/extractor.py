@@ -218,7 +218,8 @@ class GraphedExtractor:
             ats, logits = self.fused.forward_nhwc(
                 nhwc.to(self.device, torch.float32)
             )
-            return [ats], torch.softmax(logits.float(), dim=1)
+            # engine contract: fp32 taps (the bench keeps the bf16 output)
+            return [ats.float()], torch.softmax(logits.float(), dim=1)
         if self.graph is not None and xb.shape[0] == self.batch:
             prepped = self._prep(xb)
             if prepped.shape == self.static_x.shape:

@@ -152,6 +152,8 @@ class FusedPrioritizer:
                     if sa.removed_neurons
                     else None
                 )
+                if self.bf16:
+                    linv_t = linv_t.to(torch.bfloat16)
                 self.lsa_mode.append(("fused", keep, linv_t))
                 wtrains.append(xw)
                 woffs.append(woffs[-1] + xw.shape[0])
@@ -187,8 +189,12 @@ class FusedPrioritizer:
 
     def _lsa_lse(self, padded, bp, tseg_cpu, tseg):
         """Per-class whiten GEMMs + one grouped KDE launch (+ cross-rank
-        logsumexp merge of the train-shard partials in shard mode)."""
-        white = torch.zeros(bp, self.lsa_d, device=padded.device)
+        logsumexp merge of the train-shard partials in shard mode).
+
+        In bf16 mode `padded` is bf16, the whiten GEMM runs bf16 (hipBLASLt
+        MFMA) and the KDE kernel consumes the bf16 coordinates directly."""
+        wdtype = torch.bfloat16 if self.bf16 else torch.float32
+        white = torch.zeros(bp, self.lsa_d, device=padded.device, dtype=wdtype)
         for c in range(self.num_classes):
             lo, hi = int(tseg_cpu[c]), int(tseg_cpu[c + 1])
             if hi <= lo or self.lsa_mode[c][0] != "fused":
@@ -199,7 +205,7 @@ class FusedPrioritizer:
                 seg = seg.index_select(1, keep)
             white[lo:hi] = seg @ linv_t
         if self.bf16:
-            w16 = white.to(torch.bfloat16).contiguous()
+            w16 = white.contiguous()
             wf = w16.float()
             wan = (wf * wf).sum(dim=1).contiguous()
             lse = self.ext.grouped_kde_bf16(
@@ -262,8 +268,9 @@ class FusedPrioritizer:
         b = ats.shape[0]
         order, dest, tseg_cpu, tseg = self._segment(ats, pred)
         bp = int(tseg_cpu[-1])
-        padded = torch.zeros(bp, ats.shape[1], device=ats.device)
-        padded[dest] = ats[order].float()
+        pdtype = torch.bfloat16 if self.bf16 else torch.float32
+        padded = torch.zeros(bp, ats.shape[1], device=ats.device, dtype=pdtype)
+        padded[dest] = ats[order].to(pdtype)
 
         # DSA (main stream) and LSA whiten+KDE (side stream) are independent
         # given `padded`; overlap them.
@@ -280,12 +287,12 @@ class FusedPrioritizer:
                 lse.record_stream(main_stream)
 
         if self.bf16:
-            p16 = padded.to(torch.bfloat16).contiguous()
-            pf = p16.float()
+            pf = padded.float()
             an = (pf * pf).sum(dim=1).contiguous()
+            del pf
             dist, idx = self.ext.grouped_rowmin_bf16(
-                p16, self.trainS16, tseg, self.nseg, an, self.bnormS16,
-                self.jb_max,
+                padded.contiguous(), self.trainS16, tseg, self.nseg, an,
+                self.bnormS16, self.jb_max,
             )
         else:
             dist, idx = self.ext.grouped_rowmin(

@@ -117,8 +117,9 @@ class FusedResNet20:
                     variant, cur, w1.reshape(-1, 8), b1, w2.reshape(-1, 8),
                     b2, wsc.reshape(-1, 8), bsc,
                 )
-        # cur: [B, 8*8*64] NHWC bf16 — the stage-3 AT tap
-        ats = cur.float()
+        # cur: [B, 8*8*64] NHWC bf16 — the stage-3 AT tap. Returned in bf16:
+        # the bf16 pairwise kernels consume it directly (zero-copy hot
+        # path); fp32 consumers cast at their boundary.
         pooled = cur.reshape(b, 64, 64).float().mean(dim=1)  # avg over 8x8
         logits = pooled @ self.fc_w.t() + self.fc_b
-        return ats, logits
+        return cur, logits
