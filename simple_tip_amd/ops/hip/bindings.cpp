@@ -47,9 +47,6 @@ void launch_bucketize(const double*, const double*, int, int, int,
 void launch_cam_iteration(const unsigned long long*, int, int,
                           unsigned long long*, unsigned char*, long long*,
                           int*, long long*, hipStream_t);
-int launch_cam_greedy_block(const unsigned long long*, int, int,
-                            unsigned long long, unsigned char*, long long*,
-                            int*, hipStream_t);
 void launch_cam_greedy_coop(const unsigned long long*, int, int,
                             unsigned long long, unsigned char*,
                             unsigned long long*, long long*, int*, long long*,

@@ -116,75 +116,6 @@ void launch_cam_iteration(const unsigned long long* words, int rows, int W,
       part_val, part_idx, nblocks, words, W, uncovered, used, result);
 }
 
-// Whole-loop CAM in ONE kernel launch: a single 1024-thread block owns the
-// greedy loop — the uncovered mask lives in LDS, each iteration is a
-// thread-per-row popcount sweep + two-stage (wave shuffle, then cross-wave)
-// argmax, and only __syncthreads() separates iterations. The two-kernel
-// per-iteration form above cost ~200 us of dependent-launch latency per
-// pick even when hipGraph-captured (kernels themselves ~11 us); for the
-// typical ~1000-pick run that was 200+ ms of pure dispatch. One block
-// bounds parallelism at 1024 threads, but the sweep is only ~2.6 MB of
-// reads per iteration for 20k x 1000-bit profiles — L1/L2-resident.
-__launch_bounds__(1024) __global__ void cam_greedy_block_kernel(
-    const unsigned long long* __restrict__ words, int rows, int W,
-    unsigned long long init_tail_mask, unsigned char* __restrict__ used,
-    long long* __restrict__ order_out,  // [rows]
-    int* __restrict__ n_out) {
-  extern __shared__ unsigned long long sunc[];  // [W] uncovered mask
-  __shared__ long long sv[16];
-  __shared__ int si[16];
-  __shared__ int s_win;
-  __shared__ int s_cnt;
-  const int tid = threadIdx.x;
-  const int lane = lane_id();
-  const int wid = wave_id();
-  const int nwaves = blockDim.x / WAVE;
-  for (int w = tid; w < W; w += blockDim.x)
-    sunc[w] = (w == W - 1) ? init_tail_mask : ~0ull;
-  if (tid == 0) s_cnt = 0;
-  __syncthreads();
-
-  for (;;) {
-    MaxIdxLL best{-1, 0x7fffffff};
-    for (int row = tid; row < rows; row += blockDim.x) {
-      if (used[row]) continue;
-      long long c = 0;
-      const unsigned long long* r = words + (int64_t)row * W;
-      for (int w = 0; w < W; ++w) c += __popcll(r[w] & sunc[w]);
-      best = max_combine(best, MaxIdxLL{c, row});
-    }
-    // wave reduce (associative (max, min-idx) combine: order-free)
-    for (int off = 32; off >= 1; off >>= 1) {
-      MaxIdxLL o{__shfl_xor(best.v, off), __shfl_xor(best.i, off)};
-      best = max_combine(best, o);
-    }
-    if (lane == 0) {
-      sv[wid] = best.v;
-      si[wid] = best.i;
-    }
-    __syncthreads();
-    if (tid == 0) {
-      MaxIdxLL b{-1, 0x7fffffff};
-      for (int w = 0; w < nwaves; ++w)
-        b = max_combine(b, MaxIdxLL{sv[w], si[w]});
-      if (b.v > 0) {
-        order_out[s_cnt++] = b.i;
-        used[b.i] = 1;
-        s_win = b.i;
-      } else {
-        s_win = -1;
-      }
-    }
-    __syncthreads();
-    const int win = s_win;
-    if (win < 0 || s_cnt >= rows) break;
-    const unsigned long long* r = words + (int64_t)win * W;
-    for (int w = tid; w < W; w += blockDim.x) sunc[w] &= ~r[w];
-    __syncthreads();
-  }
-  if (tid == 0) *n_out = s_cnt;
-}
-
 // Persistent multi-block CAM: the single-block form above is sweep-
 // bandwidth-bound on one CU (~340 us/iteration for 20k x 1000-bit
 // profiles) and the per-iteration two-kernel form is dispatch-bound
@@ -296,22 +227,6 @@ __launch_bounds__(256) __global__ void cam_greedy_coop_kernel(
     cam_grid_barrier(barrier_ctr, &round_);
     if (*win_slot < 0 || *n_out >= rows) break;
   }
-}
-
-// Returns -1 if the single-block path is inapplicable (mask too large for
-// LDS); the caller then uses the per-iteration path.
-int launch_cam_greedy_block(const unsigned long long* words, int rows, int W,
-                            unsigned long long init_tail_mask,
-                            unsigned char* used, long long* order_out,
-                            int* n_out, hipStream_t s) {
-  const size_t lds = (size_t)W * 8;
-  if (lds > 96 * 1024) return -1;
-  auto k = cam_greedy_block_kernel;
-  hipFuncSetAttribute((const void*)k,
-                      hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
-  k<<<1, 1024, lds, s>>>(words, rows, W, init_tail_mask, used, order_out,
-                         n_out);
-  return 0;
 }
 
 void launch_cam_greedy_coop(const unsigned long long* words, int rows, int W,
