@@ -116,10 +116,11 @@ void launch_cam_iteration(const unsigned long long* words, int rows, int W,
       part_val, part_idx, nblocks, words, W, uncovered, used, result);
 }
 
-// Persistent multi-block CAM: the single-block form above is sweep-
-// bandwidth-bound on one CU (~340 us/iteration for 20k x 1000-bit
-// profiles) and the per-iteration two-kernel form is dispatch-bound
-// (~200 us/iteration). This kernel keeps CAM_NB blocks RESIDENT (1 block
+// Persistent multi-block CAM — the production path. Alternatives measured
+// and rejected (profiles/r02_cam_ladder.md): the per-iteration two-kernel
+// form above (kept as the wide-mask fallback) is dispatch-bound at
+// ~200 us/pick, and a single-block whole-loop kernel was sweep-bound on
+// one CU (~340 us/pick). This kernel keeps CAM_NB blocks RESIDENT (1 block
 // per CU is guaranteed co-residency on 256 CUs, so the software grid
 // barrier cannot deadlock) and runs the whole greedy loop with two
 // barriers per iteration: parallel sweep -> block partials -> barrier ->
