@@ -474,7 +474,13 @@ using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4b = __attribute__((ext_vector_type(4))) float;
 
 constexpr int BBK = 32;             // k per MFMA / staging step
-constexpr int BROW = BBK + 8;       // LDS halves per tile row (pad)
+// LDS halves per tile row. Must be a multiple of 8 (16-B-aligned b128
+// reads); the pad past BBK sets the bank-group stride of the fragment
+// gathers. TIP_BROW is sweepable on hardware (40 = 5-unit stride default).
+#ifndef TIP_BROW
+#define TIP_BROW 40
+#endif
+constexpr int BROW = TIP_BROW;
 
 // Stage a [128, 32] bf16 tile: thread t loads row r = t>>1, halves
 // kq = (t&1)*16 .. +15 (two 16-B units); OOB rows/ks stage zeros.
