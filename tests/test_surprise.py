@@ -162,3 +162,33 @@ def test_lsa_removes_low_variance_features():
     sa = LSA(torch.from_numpy(acts), max_features=5)
     assert 3 in sa.removed_neurons
     assert len(sa.removed_neurons) >= 5
+
+
+def test_strict_modes_flag_raises_like_reference():
+    """With strict mode on, inputs routed to a mode unseen at fit time RAISE
+    (the reference's behaviour, surprise.py:308-315) instead of scoring +inf
+    with a warning."""
+    import pytest
+
+    from simple_tip_amd.core import surprise as S
+
+    train = torch.randn(40, 6)
+    pred = torch.zeros(40, dtype=torch.long)
+    pred[20:] = 1  # classes {0, 1} at fit time
+    mm = S.MultiModalSA.build_by_class(train, pred, lambda a, p: S.MDSA(a))
+    test = torch.randn(8, 6)
+    tpred = torch.tensor([0, 1, 2, 0, 1, 2, 2, 0])  # class 2 unseen
+
+    with pytest.warns(UserWarning):
+        vals = mm(test, tpred)
+    assert torch.isinf(vals[tpred == 2]).all()
+
+    prev = S.set_strict_modes(True)
+    try:
+        with pytest.raises(ValueError):
+            mm(test, tpred)
+        dsa = S.DSA(train, pred)
+        with pytest.raises(ValueError):
+            dsa(test, tpred)
+    finally:
+        S.set_strict_modes(prev)
