@@ -504,7 +504,7 @@ def main():
     if pipelined:
         for i in range(args.warmup):
             launch_forward(i)
-            step_out = score(i)
+            score(i)
         if phase_log:
             torch.cuda.synchronize()
             step(args.warmup - 1, timed_phases=True)  # phase breakdown, unpipelined
