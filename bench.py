@@ -174,7 +174,8 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=10240, help="test inputs per GPU per step")
+    ap.add_argument("--batch", type=int, default=20480,
+                    help="test inputs per GPU per step (default = the reference's\n                    OOD test-set size; larger batches amortize per-step overhead)")
     ap.add_argument("--train-n", type=int, default=TRAIN_N)
     ap.add_argument("--setup-epochs", type=int, default=4,
                     help="untimed warm-up training epochs (class diversity)")
