@@ -71,13 +71,14 @@ static void run_pairwise(int iters) {
          ms / iters, tf);
 }
 
-static void run_resblock(int iters) {
+static void run_resblock(int iters, int variant) {
   const int batch = 4096;
-  constexpr int H = 32, W = 32, C = 16;
+  const int HH[3] = {32, 16, 8}, CC[3] = {16, 32, 64};
+  const int H = HH[variant], W = HH[variant], C = CC[variant];
   const size_t plane = (size_t)H * W * C;
   short *gin, *gout, *w1, *w2;
   float *b1, *b2;
-  constexpr int KSTEPS = (9 * C + 31) / 32;
+  const int KSTEPS = (9 * C + 31) / 32;
   CHECK(hipMalloc(&gin, batch * plane * 2));
   CHECK(hipMalloc(&gout, batch * plane * 2));
   CHECK(hipMalloc(&w1, (size_t)KSTEPS * 64 * 8 * 2));
@@ -97,20 +98,20 @@ static void run_resblock(int iters) {
   CHECK(hipMemcpy(w2, host.data(), (size_t)KSTEPS * 64 * 8 * 2, hipMemcpyHostToDevice));
   CHECK(hipMemset(b1, 0, C * 4));
   CHECK(hipMemset(b2, 0, C * 4));
-  launch_resblock(0, batch, gin, gout, w1, b1, w2, b2, 0);
+  launch_resblock(variant, batch, gin, gout, w1, b1, w2, b2, 0);
   CHECK(hipDeviceSynchronize());
   hipEvent_t t0, t1;
   hipEventCreate(&t0);
   hipEventCreate(&t1);
   hipEventRecord(t0);
   for (int i = 0; i < iters; ++i)
-    launch_resblock(0, batch, gin, gout, w1, b1, w2, b2, 0);
+    launch_resblock(variant, batch, gin, gout, w1, b1, w2, b2, 0);
   hipEventRecord(t1);
   CHECK(hipDeviceSynchronize());
   float ms;
   hipEventElapsedTime(&ms, t0, t1);
-  const double flops = 2.0 * batch * 2 * 1024 * 16 * 144;
-  printf("resblock<32,32,16> b=%d: %.3f ms/iter, %.1f TF\n", batch,
+  const double flops = 2.0 * batch * 2 * (double)(H * W) * C * (9 * C);
+  printf("resblock<%d,%d,%d> b=%d: %.3f ms/iter, %.1f TF\n", H, W, C, batch,
          ms / iters, flops * iters / (ms / 1e3) / 1e12);
 }
 
@@ -437,6 +438,6 @@ int main(int argc, char** argv) {
     run_verify(argc > 2 ? atoi(argv[2]) : 1024, argc > 3 ? atoi(argv[3]) : 1500,
                argc > 4 ? atoi(argv[4]) : 4096);
   else
-    run_resblock(iters);
+    run_resblock(iters, argc > 3 ? atoi(argv[3]) : 0);
   return 0;
 }

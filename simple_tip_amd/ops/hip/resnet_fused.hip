@@ -140,6 +140,77 @@ TIP_DEV void conv3x3(
   const int j = lane & 15;        // output channel within tile
   const int g = lane >> 4;        // k-group (8 consecutive k)
 
+  // Small-plane specialization (PIX_TILES <= 4, i.e. the 8x8 stage): each
+  // wave owns at most ONE pixel tile, so the generic path's dual-PIXEL
+  // MFMA chains collapse to a single dependent chain (measured 24% slower
+  // per FLOP than the 32x32 stage). Interleave TWO COUT tiles instead:
+  // the A fragment is shared, the B fragments load from L2 per k-step
+  // (high occupancy here — 6 blocks/CU — hides the load latency the
+  // preload design was protecting the low-occupancy stages from). The
+  // per-(ct, ks) accumulation order is unchanged, so results are bitwise
+  // identical to the generic path.
+  if constexpr (PIX_TILES <= 4 && (COUT_TILES % 2) == 0) {
+    const int pt = wid;
+    if (pt >= PIX_TILES) return;
+    const int apix = pt * 16 + j;
+    const int aoy = apix / OW, aox = apix - aoy * OW;
+    const int abase = In::unit(aoy * STRIDE, aox * STRIDE, 0);
+    for (int ct = 0; ct < COUT_TILES; ct += 2) {
+      const short* wp0 = wpack + ((int64_t)ct * KSTEPS) * 64 * 8 + lane * 8;
+      const short* wp1 = wp0 + (int64_t)KSTEPS * 64 * 8;
+      f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+      f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        const int k0 = ks * 32 + g * 8;
+        short8 a;
+        if (k0 < K) {
+          const int tap = k0 / C;
+          const int ci = k0 & (C - 1);
+          const int dy = (tap * 11) >> 5;
+          const int dx = tap - dy * 3;
+          a = lds_read_unit(in_lds, abase + In::unit(dy, dx, ci >> 3));
+        } else {
+          a = short8{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+        const short8 b0 =
+            *reinterpret_cast<const short8*>(wp0 + (int64_t)ks * 64 * 8);
+        const short8 b1 =
+            *reinterpret_cast<const short8*>(wp1 + (int64_t)ks * 64 * 8);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
+      }
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const int cout = (ct + half) * 16 + j;
+        const float bs = bias[cout];
+        const f32x4& acc = half ? acc1 : acc0;
+        const int pix0 = pt * 16 + g * 4;
+        const int oy = pix0 / OW, ox = pix0 - oy * OW;
+        const int u0 = Out::unit(oy + 1, ox + 1, cout >> 3);
+        const int e = cout & 7;
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          float v = acc[reg] + bs;
+          if (RESID) {
+            const short* runit = rlds + swz(u0 + reg * Out::XSTEP) * 8;
+            v += __bfloat162float(reinterpret_cast<const bf16*>(runit)[e]);
+          }
+          v = fmaxf(v, 0.f);
+          const bf16 ov = __float2bfloat16(v);
+          if (TO_LDS) {
+            short* unit = out_lds + swz(u0 + reg * Out::XSTEP) * 8;
+            reinterpret_cast<bf16*>(unit)[e] = ov;
+          } else {
+            reinterpret_cast<bf16*>(gout)[(int64_t)(pix0 + reg) * COUT + cout] =
+                ov;
+          }
+        }
+      }
+    }
+    return;
+  }
+
   // cout-tile outer loop: the tile's weight fragments are preloaded into
   // registers ONCE (KSTEPS x 16 B per lane) so the MFMA loop is pure
   // ds_read + mfma — no global loads on the critical path (the per-mfma
