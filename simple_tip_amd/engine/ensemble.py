@@ -40,6 +40,16 @@ def load_model(study_name: str, model_id: int, factory: Callable[[], torch.nn.Mo
     return model
 
 
+def _run_on_assigned_gpu(task: Callable[[int], Any], ngpu: int, mid: int):
+    """Pool target: pin this (fresh, spawned) worker to GPU ``mid % ngpu``
+    BEFORE torch initialises HIP — model ids spread round-robin over the
+    node's GPUs, which is the natural placement for the reference's
+    "100 independent models" experiment shape on an 8-GPU node."""
+    if ngpu > 1 and "HIP_VISIBLE_DEVICES" not in os.environ:
+        os.environ["HIP_VISIBLE_DEVICES"] = str(mid % ngpu)
+    return task(mid)
+
+
 def run_tasks(
     task: Callable[[int], Any],
     model_ids: List[int],
@@ -50,11 +60,15 @@ def run_tasks(
     num_processes == 0 runs inline (tests, single-GPU boxes); otherwise a
     spawn pool with maxtasksperchild=1 (one task per process, then exit —
     the reference's TF memory-leak workaround, kept because each child also
-    gets a fresh HIP context).
+    gets a fresh HIP context). On a multi-GPU node each model id is pinned
+    to GPU ``id % num_gpus`` (round-robin model-level parallelism).
     ``task`` must be picklable (module-level function / functools.partial).
     """
     if num_processes <= 0 or len(model_ids) <= 1:
         return [task(mid) for mid in model_ids]
+    from functools import partial
+
+    ngpu = torch.cuda.device_count() if torch.cuda.is_available() else 0
     ctx = mp.get_context("spawn")
     with ctx.Pool(processes=num_processes, maxtasksperchild=1) as pool:
-        return pool.map(task, model_ids)
+        return pool.map(partial(_run_on_assigned_gpu, task, ngpu), model_ids)

@@ -193,6 +193,45 @@ def test_world2_coverage_worker_sharded_matches_dense():
     _run_world(_coverage_worker, 2, 14, ())
 
 
+def _uneven_worker(rank, world, port, q):
+    """Uneven shards (26 inputs over 3 ranks) with batch boundaries that do
+    NOT line up with the shard edges: scores must still match the dense run
+    (popcounts are exact; only batch regrouping changes)."""
+    try:
+        _init(rank, world, port)
+        import torch.distributed as dist
+
+        from simple_tip_amd.engine.coverage_handler import CoverageWorker
+        from simple_tip_amd.engine.model_handler import BaseModel
+        from simple_tip_amd.models import MnistCNN
+
+        torch.manual_seed(0)
+        model = MnistCNN()
+        rng = np.random.RandomState(5)
+        train = rng.rand(29, 1, 28, 28).astype(np.float32)
+        test = rng.rand(26, 1, 28, 28).astype(np.float32)
+
+        def build(ds):
+            return CoverageWorker(
+                BaseModel(model, [0, 3], predict_batch=4), train, dist_shard=ds
+            )
+
+        _, s_d, c_d = build(False).evaluate_all(test, "nominal")
+        _, s_s, c_s = build(True).evaluate_all(test, "nominal")
+        for m in s_d:
+            assert np.array_equal(s_d[m], s_s[m]), f"{m} scores mismatch"
+            assert c_d[m] == c_s[m], f"{m} cam order mismatch"
+
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"fail: {e!r}"))
+
+
+def test_world3_coverage_worker_uneven_shards():
+    _run_world(_uneven_worker, 3, 17, ())
+
+
 # ---------------------------------------------------------------------------
 # Engine level: full eval_prioritization, world-4 == world-1
 # ---------------------------------------------------------------------------
