@@ -421,6 +421,45 @@ static void run_grouped_bf16(int iters, int perclass_test, int perclass_train,
          C, bp, perclass_train, k, ms / iters, tf, dsum, isum);
 }
 
+static void run_downblock(int iters, int variant) {
+  const int batch = 4096;
+  const int HH[2] = {32, 16}, CC[2] = {16, 32};
+  const int H = HH[variant], W = HH[variant], C = CC[variant];
+  const int OH = H / 2, OW = W / 2, C2 = 2 * C;
+  short *gin, *gout, *w1, *w2, *wsc;
+  float *b1, *b2, *bsc;
+  const int KS1 = (9 * C + 31) / 32, KS2 = (9 * C2 + 31) / 32,
+            KSC = (C + 31) / 32;
+  CHECK(hipMalloc(&gin, (size_t)batch * H * W * C * 2));
+  CHECK(hipMalloc(&gout, (size_t)batch * OH * OW * C2 * 2));
+  CHECK(hipMalloc(&w1, (size_t)(C2 / 16) * KS1 * 64 * 8 * 2));
+  CHECK(hipMalloc(&w2, (size_t)(C2 / 16) * KS2 * 64 * 8 * 2));
+  CHECK(hipMalloc(&wsc, (size_t)(C2 / 16) * KSC * 64 * 8 * 2));
+  CHECK(hipMalloc(&b1, C2 * 4));
+  CHECK(hipMalloc(&b2, C2 * 4));
+  CHECK(hipMalloc(&bsc, C2 * 4));
+  CHECK(hipMemset(b1, 0, C2 * 4));
+  CHECK(hipMemset(b2, 0, C2 * 4));
+  CHECK(hipMemset(bsc, 0, C2 * 4));
+  launch_downblock(variant, batch, gin, gout, w1, b1, w2, b2, wsc, bsc, 0);
+  CHECK(hipDeviceSynchronize());
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0);
+  hipEventCreate(&t1);
+  hipEventRecord(t0);
+  for (int i = 0; i < iters; ++i)
+    launch_downblock(variant, batch, gin, gout, w1, b1, w2, b2, wsc, bsc, 0);
+  hipEventRecord(t1);
+  CHECK(hipDeviceSynchronize());
+  float ms;
+  hipEventElapsedTime(&ms, t0, t1);
+  // conv1 (stride2: OHxOW out, K=9C) + conv2 (K=9*C2) + 1x1 shortcut
+  const double flops =
+      2.0 * batch * (double)(OH * OW) * C2 * (9.0 * C + 9.0 * C2 + C);
+  printf("downblock<%d,%d,%d> b=%d: %.3f ms/iter, %.1f TF\n", H, W, C,
+         batch, ms / iters, flops * iters / (ms / 1e3) / 1e12);
+}
+
 int main(int argc, char** argv) {
   const char* which = argc > 1 ? argv[1] : "pairwise";
   const int iters = argc > 2 ? atoi(argv[2]) : 10;
@@ -437,6 +476,8 @@ int main(int argc, char** argv) {
   else if (!strcmp(which, "verify"))
     run_verify(argc > 2 ? atoi(argv[2]) : 1024, argc > 3 ? atoi(argv[3]) : 1500,
                argc > 4 ? atoi(argv[4]) : 4096);
+  else if (!strcmp(which, "downblock"))
+    run_downblock(iters, argc > 3 ? atoi(argv[3]) : 0);
   else
     run_resblock(iters, argc > 3 ? atoi(argv[3]) : 0);
   return 0;
