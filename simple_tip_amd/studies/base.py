@@ -70,12 +70,23 @@ def train_classifier(
         xt = xt.float()
     yt = torch.as_tensor(np.asarray(y).reshape(-1)).long()
     use_amp = device.type == "cuda"
+    # NHWC training on GPU for conv models: bf16-autocast NCHW convs hit
+    # MIOpen's NAIVE weight-gradient fallback (~30 ms/launch measured in
+    # bench traces); channels_last routes the backward to the igemm path.
+    nhwc = use_amp and xt.dim() == 4
+    if nhwc:
+        model.to(memory_format=torch.channels_last)
     for epoch in range(epochs):
         # identical permutation on every rank, rank-strided shard of it
         gen = torch.Generator().manual_seed((seed or 0) * 1000 + epoch)
         perm = torch.randperm(n_train, generator=gen)
         rank = pdist.get_rank()
-        total, correct, loss_sum = 0, 0, 0.0
+        total = 0
+        # epoch statistics accumulate ON DEVICE: a float()/int() per step is
+        # a host sync, and at reference batch sizes (cifar10: batch 32,
+        # 1.5k steps/epoch) the syncs dominated the epoch wall time
+        loss_acc = torch.zeros((), dtype=torch.float64, device=device)
+        corr_acc = torch.zeros((), dtype=torch.int64, device=device)
         total_steps = max(1, (n_train + batch_size - 1) // batch_size)
         # every rank runs the same number of steps (DDP all-reduce must
         # match); overhanging ranks wrap to batch 0
@@ -86,18 +97,20 @@ def train_classifier(
                 idx = perm[0:batch_size]
             xb = xt[idx].to(device, non_blocking=True)
             yb = yt[idx].to(device, non_blocking=True)
+            if nhwc:
+                xb = xb.to(memory_format=torch.channels_last)
             opt.zero_grad(set_to_none=True)
             with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_amp):
                 logits = train_model(xb)
                 loss = loss_fn(logits.float(), yb)
             loss.backward()
             opt.step()
-            loss_sum += float(loss.detach()) * len(idx)
-            correct += int((logits.argmax(dim=1) == yb).sum())
+            loss_acc += loss.detach().double() * len(idx)
+            corr_acc += (logits.argmax(dim=1) == yb).sum()
             total += len(idx)
         logger.info(
             "epoch %d/%d loss %.4f acc %.3f", epoch + 1, epochs,
-            loss_sum / max(total, 1), correct / max(total, 1),
+            float(loss_acc) / max(total, 1), int(corr_acc) / max(total, 1),
         )
     model.eval()
     return model
