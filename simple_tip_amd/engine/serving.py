@@ -9,8 +9,11 @@ kernels: the batch is class-sorted once, padded to 128-row segments, and a
 SINGLE kernel launch covers every class for DSA (and one for LSA), cutting
 the host op count per batch to ~40.
 
-Numerics are identical to the per-class path (same kernels, same fp32
-accumulation, same tie rules); tests/test_gpu_serving.py asserts equality.
+Numerics: the fp32 mode is identical to the per-class path (same kernels,
+same fp32 accumulation, same tie rules; tests/test_gpu_serving.py asserts
+equality). The optional bf16 mode (``pairwise_dtype=torch.bfloat16``) uses
+the bf16-operand MFMA kernels — 4.6x faster, scores shift only by operand
+rounding (rank correlation > 0.99 vs fp32, pinned by test).
 """
 
 from typing import Optional, Tuple
