@@ -232,6 +232,52 @@ def test_world3_coverage_worker_uneven_shards():
     _run_world(_uneven_worker, 3, 17, ())
 
 
+def _surprise_uneven_worker(rank, world, port, q):
+    """Input-sharded SurpriseHandler with shard sizes that do not align
+    with predict_batch: all-gathered SA scores must match the dense run
+    (forward batch regrouping shifts GEMM rounding, so allclose)."""
+    try:
+        _init(rank, world, port)
+        import torch.distributed as dist
+
+        from simple_tip_amd.engine.surprise_handler import SurpriseHandler
+        from simple_tip_amd.models import MnistCNN
+
+        torch.manual_seed(0)
+        np.random.seed(0)  # pin pc-mlsa EM init on the dense run too
+        model = MnistCNN()
+        rng = np.random.RandomState(9)
+        train = rng.rand(34, 1, 28, 28).astype(np.float32)
+        nominal = rng.rand(17, 1, 28, 28).astype(np.float32)
+
+        def run(ds):
+            np.random.seed(0)
+            sh = SurpriseHandler(
+                model, sa_layers=[3], training_dataset=train,
+                predict_batch=8, dist_shard=ds,
+            )
+            return sh.evaluate_all({"nominal": nominal})
+
+        dense = run(False)
+        shard = run(True)
+        for sa in dense:
+            sd, od, _ = dense[sa]["nominal"]
+            ss, os_, _ = shard[sa]["nominal"]
+            fin = np.isfinite(sd)
+            assert np.array_equal(fin, np.isfinite(ss)), sa
+            assert np.allclose(sd[fin], ss[fin], rtol=1e-6, atol=1e-8), sa
+            assert np.array_equal(od, os_), f"{sa} cam order"
+
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"fail: {e!r}"))
+
+
+def test_world3_surprise_handler_uneven_shards():
+    _run_world(_surprise_uneven_worker, 3, 18, ())
+
+
 # ---------------------------------------------------------------------------
 # Engine level: full eval_prioritization, world-4 == world-1
 # ---------------------------------------------------------------------------
