@@ -20,8 +20,6 @@ import os
 import sys
 import time
 
-import numpy as np
-
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 D, NTRAIN, BATCH, CLASSES = 512, 20000, 4096, 10

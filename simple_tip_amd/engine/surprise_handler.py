@@ -16,7 +16,7 @@ from ..config import NUM_SC_BUCKETS
 from ..core.prioritizers import cam
 from ..core.surprise import DSA, LSA, MDSA, MLSA, MultiModalSA, SurpriseCoverageMapper
 from ..core.timer import DeviceTimer
-from .model_handler import BaseModel, iter_batches
+from .model_handler import BaseModel
 
 logger = logging.getLogger(__name__)
 

@@ -18,7 +18,6 @@ from .. import ops
 from .dist import (
     _staging_device,
     gather_tensors,
-    get_rank,
     get_world_size,
     is_initialized,
     shard_slice,

@@ -9,7 +9,7 @@ training runs in bf16 autocast on the GPU.
 import logging
 import os
 from functools import partial
-from typing import Callable, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 import torch
