@@ -278,6 +278,45 @@ def test_world3_surprise_handler_uneven_shards():
     _run_world(_surprise_uneven_worker, 3, 18, ())
 
 
+def _imdb_coverage_worker(rank, world, port, q):
+    """BASELINE config 4's exact combination: the IMDB transformer's
+    neuron-coverage profiles sharded over inputs and reassembled with the
+    coverage-bitmap OR all-reduce."""
+    try:
+        _init(rank, world, port)
+        import torch.distributed as dist
+
+        from simple_tip_amd.engine.coverage_handler import CoverageWorker
+        from simple_tip_amd.engine.model_handler import BaseModel
+        from simple_tip_amd.models.transformer import ImdbTransformer
+        from simple_tip_amd.studies.synthetic import synthetic_tokens
+
+        torch.manual_seed(0)
+        model = ImdbTransformer()
+        tx, _ = synthetic_tokens("cfg4", "train", 30, 100, 2000, 2)
+        nx, _ = synthetic_tokens("cfg4", "test", 18, 100, 2000, 2)
+
+        def build(ds):
+            return CoverageWorker(
+                BaseModel(model, [3, 5], predict_batch=4), tx, dist_shard=ds
+            )
+
+        _, s_d, c_d = build(False).evaluate_all(nx, "nominal")
+        _, s_s, c_s = build(True).evaluate_all(nx, "nominal")
+        for m in s_d:
+            assert np.array_equal(s_d[m], s_s[m]), f"{m} scores mismatch"
+            assert c_d[m] == c_s[m], f"{m} cam order mismatch"
+
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, f"fail: {e!r}"))
+
+
+def test_world2_imdb_coverage_bitmap_allreduce():
+    _run_world(_imdb_coverage_worker, 2, 19, ())
+
+
 # ---------------------------------------------------------------------------
 # Engine level: full eval_prioritization, world-4 == world-1
 # ---------------------------------------------------------------------------
