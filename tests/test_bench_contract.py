@@ -52,3 +52,31 @@ def test_bench_gpus_flag_validated():
     )
     assert out.returncode != 0
     assert "WORLD_SIZE" in (out.stderr + out.stdout)
+
+
+@pytest.mark.timeout(900)
+def test_bench_shard_train_world2_contract():
+    """The strong-scaling mode end-to-end on CPU (gloo world 2): sharded
+    forward + train-AT-sharded scoring must produce the contract JSON with
+    scaling=strong and the whole-job (not per-rank) aggregate."""
+    import numpy as np
+
+    port = int(np.random.RandomState(os.getpid()).randint(20000, 40000))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+            "--master-port", str(port),
+            os.path.join(REPO, "bench.py"), "--gpus", "2",
+            "--steps", "2", "--warmup", "1", "--batch", "128",
+            "--train-n", "512", "--setup-epochs", "1", "--shard-train",
+        ],
+        capture_output=True, text=True, timeout=840, cwd=REPO,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.strip().splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["scaling"] == "strong"
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "train-shard2"
+    assert d["config"]["global_batch"] == 128  # total work fixed, not x2
