@@ -231,17 +231,10 @@ class FusedPrioritizer:
         """Strict-less rank-ordered merge of (min, global argmin) partials —
         lowest rank holds the lowest global rows per class, so ties keep the
         single-device lowest-index rule."""
-        from ..parallel.dist import gather_tensors, get_world_size
+        from ..parallel.dist import gather_tensors
+        from ..parallel.sharded import fold_rowmin_partials
 
-        world = get_world_size()
-        dg = gather_tensors(dist)
-        ig = gather_tensors(idx)
-        best_d, best_i = dg[0], ig[0]
-        for r in range(1, world):
-            take = dg[r] < best_d
-            best_d = torch.where(take, dg[r], best_d)
-            best_i = torch.where(take, ig[r], best_i)
-        return best_d, best_i
+        return fold_rowmin_partials(gather_tensors(dist), gather_tensors(idx))
 
     def _segment(self, ats: torch.Tensor, pred: torch.Tensor):
         """Class-sort + pad to 128-row segments."""

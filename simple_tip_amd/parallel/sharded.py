@@ -31,6 +31,20 @@ def shard_rows(full: torch.Tensor) -> Tuple[torch.Tensor, int]:
     return full[s].contiguous(), s.start
 
 
+def fold_rowmin_partials(dists, idxs) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Rank-ordered strict-less fold of per-shard (min, GLOBAL argmin)
+    partials. With shards holding ascending global row ranges, strict-less
+    in rank order preserves the single-device lowest-index tie rule —
+    the determinism property tests/test_sharded.py pins with hypothesis.
+    """
+    best_d, best_i = dists[0], idxs[0]
+    for r in range(1, len(dists)):
+        take = dists[r] < best_d
+        best_d = torch.where(take, dists[r], best_d)
+        best_i = torch.where(take, idxs[r], best_i)
+    return best_d, best_i
+
+
 def sharded_rowmin_l2(
     test: torch.Tensor,
     train_shard: torch.Tensor,
@@ -39,10 +53,9 @@ def sharded_rowmin_l2(
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Global (min L2 distance, global argmin) with the train rows sharded.
 
-    Partials merge in rank order with strict-less comparison, preserving the
-    lowest-global-index tie rule of the single-device path. A rank whose
-    shard is empty (class smaller than the world size) contributes +inf
-    partials and never wins the merge.
+    Partials merge with :func:`fold_rowmin_partials`. A rank whose shard is
+    empty (class smaller than the world size) contributes +inf partials and
+    never wins the merge.
     """
     if train_shard.shape[0] == 0:
         d = torch.full(
@@ -54,16 +67,7 @@ def sharded_rowmin_l2(
         i = i + shard_offset
     if not is_initialized():
         return d, i
-    world = get_world_size()
-    dg = gather_tensors(d)
-    ig = gather_tensors(i)
-    best_d, best_i = dg[0], ig[0]
-    for r in range(1, world):
-        # ranks hold ascending global offsets: strict less keeps lowest idx
-        take = dg[r] < best_d
-        best_d = torch.where(take, dg[r], best_d)
-        best_i = torch.where(take, ig[r], best_i)
-    return best_d, best_i
+    return fold_rowmin_partials(gather_tensors(d), gather_tensors(i))
 
 
 def sharded_kde_logsumexp(

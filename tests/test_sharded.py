@@ -133,3 +133,49 @@ def test_gloo_world2_ddp_training():
         p.join(timeout=60)
     for rank, status in results:
         assert status == "ok", f"rank {rank}: {status}"
+
+
+def test_fold_rowmin_partials_property():
+    """hypothesis: for ANY split of the train rows into contiguous shards,
+    the rank-ordered fold of per-shard (min, global argmin) equals the dense
+    single-pass result — including the lowest-index tie rule."""
+    from hypothesis import given, settings, strategies as st
+
+    from simple_tip_amd.parallel.sharded import fold_rowmin_partials
+
+    @settings(max_examples=60, deadline=None)
+    @given(
+        n=st.integers(4, 60),
+        m=st.integers(1, 12),
+        world=st.integers(1, 5),
+        seed=st.integers(0, 10_000),
+        quantize=st.booleans(),
+    )
+    def check(n, m, world, seed, quantize):
+        g = torch.Generator().manual_seed(seed)
+        test = torch.randn(m, 4, generator=g).double()
+        train = torch.randn(n, 4, generator=g).double()
+        if quantize:  # force exact distance ties to exercise the tie rule
+            test = test.round()
+            train = train.round()
+        dense_d, dense_i = ops.rowmin_l2(test, train)
+        # split into `world` contiguous shards (possibly empty)
+        cuts = sorted(
+            int(x) for x in torch.randint(0, n + 1, (world - 1,), generator=g)
+        )
+        bounds = [0] + cuts + [n]
+        dists, idxs = [], []
+        for r in range(world):
+            lo, hi = bounds[r], bounds[r + 1]
+            if hi == lo:
+                dists.append(torch.full((m,), float("inf")).double())
+                idxs.append(torch.zeros(m, dtype=torch.int64))
+            else:
+                d, i = ops.rowmin_l2(test, train[lo:hi])
+                dists.append(d)
+                idxs.append(i + lo)
+        fd, fi = fold_rowmin_partials(dists, idxs)
+        assert torch.equal(fd, dense_d)
+        assert torch.equal(fi, dense_i)
+
+    check()
